@@ -1,0 +1,122 @@
+/* spectre_gpu.h — C ABI of the MI355X-native BN254 MSM/NTT proving backend.
+ *
+ * This is the drop-in boundary for the Spectre/halo2 proving hot path: the
+ * reference's own seam is the pair of free functions
+ *   best_multiexp(coeffs: &[Fr], bases: &[G1Affine]) -> G1   and
+ *   best_fft(a: &mut [Fr], omega: Fr, log_n: u32)
+ * in halo2_proofs::arithmetic / halo2curves (reached from every proof and
+ * keygen via lightclient-circuits/src/util/circuit.rs:131,158,177,211 in the
+ * reference tree). A patched halo2_proofs calls these entry points through a
+ * ~100-line unsafe extern "C" block; see INTEGRATION.md for the Rust-side
+ * binding a maintainer would add.
+ *
+ * Data formats (= halo2curves-axiom 0.5.2 memory images, little-endian):
+ *   Fr/Fq element : 32 B = 4 x u64 LE limbs of the Montgomery residue
+ *                   a*2^256 mod m (the raw &[Fr] slice memory).
+ *   canonical Fr  : 32 LE bytes of the integer itself (Fr::to_repr()).
+ *   G1 affine     : 64 B = x || y Montgomery Fq; identity = 64 zero bytes.
+ *   G1 Jacobian   : 96 B = X || Y || Z Montgomery Fq; identity: Z = 0
+ *                   (shard partial sums only; never a caller-visible result).
+ *
+ * Thread safety: all calls on one ctx are serialized internally; halo2 may
+ * invoke commits from concurrent rayon workers. Errors: 0 = ok, negative =
+ * failure; spectre_gpu_last_error() returns a thread-local message. The
+ * library REQUIRES a visible AMD GPU for every compute entry point (there is
+ * deliberately no CPU fallback — a missing GPU fails loudly); the only
+ * host-only entry points are spectre_gpu_msm_g1_combine and the
+ * format/version queries.
+ */
+#ifndef SPECTRE_GPU_H
+#define SPECTRE_GPU_H
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct spectre_gpu_ctx spectre_gpu_ctx;
+
+/* scalars argument encoding for MSM entry points */
+#define SPECTRE_SCALARS_MONTGOMERY 0u /* raw &[Fr] memory (the real seam) */
+#define SPECTRE_SCALARS_CANONICAL 1u  /* already to_repr()'d */
+
+#define SPECTRE_MSM_NUM_WINDOWS 16u /* 16-bit signed windows over 256 bits */
+
+/* ctx over `device_count` HIP devices (device_ids NULL -> 0..count-1).
+ * Returns NULL on failure (no GPU, bad ids). */
+spectre_gpu_ctx* spectre_gpu_init(int device_count, const int* device_ids);
+void spectre_gpu_destroy(spectre_gpu_ctx*);
+const char* spectre_gpu_last_error(void); /* thread-local, valid until next call */
+int spectre_gpu_device_count(spectre_gpu_ctx*);
+const char* spectre_gpu_version(void);
+
+/* ---- MSM (host pointers) ------------------------------------------------
+ * out_affine = Sum_i scalars[i] * bases[i].  bases_id != 0: the uploaded
+ * bases are cached on-device under (bases_id, n) and `bases` may be NULL on
+ * subsequent calls (SRS reuse across proofs). num_gpus in [1, device_count]:
+ * the scalar/base stream is sharded contiguously, each device runs the full
+ * Pippenger pipeline on its shard, partial window sums are combined on host
+ * in rank order. */
+int spectre_gpu_msm_g1(spectre_gpu_ctx*, uint64_t bases_id,
+                       const uint8_t* bases /* n*64B */,
+                       const uint8_t* scalars /* n*32B */, uint64_t n,
+                       uint32_t flags, int num_gpus, uint8_t out_affine[64]);
+
+/* ---- MSM (device-resident, single device `dev`) ------------------------ */
+int spectre_gpu_msm_g1_device(spectre_gpu_ctx*, int dev,
+                              const void* d_bases, const void* d_scalars,
+                              uint64_t n, uint32_t flags,
+                              uint8_t out_affine[64]);
+/* Per-window Jacobian partial sums of this shard (for cross-process
+ * sharding, e.g. one rank per GPU exchanging partials over RCCL):
+ * out_partials = SPECTRE_MSM_NUM_WINDOWS * 96 bytes. */
+int spectre_gpu_msm_g1_shard_device(spectre_gpu_ctx*, int dev,
+                                    const void* d_bases,
+                                    const void* d_scalars, uint64_t n,
+                                    uint32_t flags, uint8_t* out_partials);
+/* As shard_device, plus per-stage timings from HIP events recorded on the
+ * library's own stream (the bench's live roofline measurement):
+ * out_ms[0..5] = digits, radix-sort, offsets, bucket-accumulate, chunks,
+ * reduce-tree; out_ms[6] = total GPU ms; out_ms[7] = count of real
+ * (non-zero-digit) sort entries processed. */
+int spectre_gpu_msm_g1_shard_device_timed(spectre_gpu_ctx*, int dev,
+                                          const void* d_bases,
+                                          const void* d_scalars, uint64_t n,
+                                          uint32_t flags,
+                                          uint8_t* out_partials,
+                                          double out_ms[8]);
+/* Combine gathered shard partials (host-only, deterministic rank order):
+ * partials = nshards * NUM_WINDOWS * 96 B. */
+int spectre_gpu_msm_g1_combine(const uint8_t* partials, uint32_t nshards,
+                               uint8_t out_affine[64]);
+
+/* ---- NTT ----------------------------------------------------------------
+ * In-place radix-2 NTT over Fr, Montgomery-form data, exactly halo2's
+ * best_fft / EvaluationDomain semantics:
+ *   if coset_gen && !inverse : data[i] *= coset_gen^i            (before)
+ *   data <- DFT(data, omega)     [caller passes omega_inv to invert]
+ *   if inverse               : data[i] *= n^{-1}
+ *   if coset_gen && inverse  : data[i] *= coset_gen^i   (after; pass g^{-1})
+ * log_n <= 24 (2-adicity headroom and scratch sizing; the reference needs
+ * 2^20..2^24). */
+int spectre_gpu_ntt_fr(spectre_gpu_ctx*, uint8_t* data, uint32_t log_n,
+                       const uint8_t omega[32], int inverse,
+                       const uint8_t* coset_gen);
+int spectre_gpu_ntt_fr_device(spectre_gpu_ctx*, int dev, void* d_data,
+                              uint32_t log_n, const uint8_t omega[32],
+                              int inverse, const uint8_t* coset_gen);
+
+/* ---- device memory helpers (for C callers and the bench harness) ------- */
+int spectre_gpu_malloc(spectre_gpu_ctx*, int dev, size_t bytes, void** d_ptr);
+int spectre_gpu_free(spectre_gpu_ctx*, int dev, void* d_ptr);
+int spectre_gpu_upload(spectre_gpu_ctx*, int dev, void* d_dst, const void* src,
+                       size_t bytes);
+int spectre_gpu_download(spectre_gpu_ctx*, int dev, void* dst,
+                         const void* d_src, size_t bytes);
+int spectre_gpu_synchronize(spectre_gpu_ctx*, int dev);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* SPECTRE_GPU_H */

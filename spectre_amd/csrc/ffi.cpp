@@ -1,0 +1,323 @@
+// ffi.cpp — C-ABI implementation of libspectre_gpu.so (see
+// include/spectre_gpu.h for the boundary contract and INTEGRATION.md for the
+// Rust-side binding that slots this beneath halo2's best_multiexp/best_fft).
+//
+// Host-side work is deliberately tiny: context/stream/cache management,
+// SRS upload caching, and the final window-Horner (<= 256 point ops per MSM)
+// + one field inversion to normalize to affine. All bulk compute runs in the
+// HIP kernels (msm.hip / ntt.hip); there is no CPU fallback path.
+#include "internal.hpp"
+#include "spectre_gpu.h"
+#include <cstdarg>
+#include <cstdio>
+
+thread_local std::string g_last_error;
+void set_err(const char* fmt, ...) {
+    char buf[512];
+    va_list ap;
+    va_start(ap, fmt);
+    vsnprintf(buf, sizeof buf, fmt, ap);
+    va_end(ap);
+    g_last_error = buf;
+}
+
+extern "C" {
+
+const char* spectre_gpu_last_error(void) { return g_last_error.c_str(); }
+const char* spectre_gpu_version(void) { return "spectre-amd 0.1.0 gfx950"; }
+
+spectre_gpu_ctx* spectre_gpu_init(int device_count, const int* device_ids) {
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0) {
+        set_err("spectre_gpu_init: no visible HIP devices (this library has "
+                "no CPU fallback by design)");
+        return nullptr;
+    }
+    if (device_count <= 0) device_count = 1;
+    auto* ctx = new spectre_gpu_ctx();
+    for (int i = 0; i < device_count; i++) {
+        int id = device_ids ? device_ids[i] : i;
+        if (id < 0 || id >= ndev) {
+            set_err("spectre_gpu_init: device id %d out of range (%d devices)",
+                    id, ndev);
+            delete ctx;
+            return nullptr;
+        }
+        DeviceState ds;
+        ds.device_id = id;
+        if (hipSetDevice(id) != hipSuccess ||
+            hipStreamCreate(&ds.stream) != hipSuccess) {
+            set_err("spectre_gpu_init: failed to init device %d", id);
+            delete ctx;
+            return nullptr;
+        }
+        ctx->devs.push_back(ds);
+    }
+    return ctx;
+}
+
+void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
+    if (!ctx) return;
+    for (auto& ds : ctx->devs) {
+        (void)hipSetDevice(ds.device_id);
+        for (void* p :
+             {(void*)ds.d_keys_in, (void*)ds.d_keys_out, (void*)ds.d_vals_in,
+              (void*)ds.d_vals_out, (void*)ds.d_sort_tmp, (void*)ds.d_offsets,
+              (void*)ds.d_buckets, (void*)ds.d_red, (void*)ds.d_scalars,
+              (void*)ds.d_bases, (void*)ds.d_ntt_tmp, (void*)ds.d_cosetA})
+            if (p) (void)hipFree(p);
+        for (auto& kv : ds.bases_cache)
+            if (kv.second.d_ptr) (void)hipFree(kv.second.d_ptr);
+        for (auto& kv : ds.plans) {
+            if (kv.second.tw1) (void)hipFree(kv.second.tw1);
+            if (kv.second.tw2) (void)hipFree(kv.second.tw2);
+            if (kv.second.twB) (void)hipFree(kv.second.twB);
+        }
+        if (ds.stream) (void)hipStreamDestroy(ds.stream);
+    }
+    delete ctx;
+}
+
+int spectre_gpu_device_count(spectre_gpu_ctx* ctx) {
+    return ctx ? (int)ctx->devs.size() : 0;
+}
+
+// ---------------------------------------------------------------- helpers
+static int check_dev(spectre_gpu_ctx* ctx, int dev) {
+    if (!ctx || dev < 0 || dev >= (int)ctx->devs.size()) {
+        set_err("invalid ctx/device index %d", dev);
+        return -1;
+    }
+    return 0;
+}
+
+// Horner over the 16 windows + normalize to affine (host; <=256 point ops).
+static void winsums_to_affine(const g1_jac* wins, uint8_t out[64]) {
+    g1_jac res;
+    g1j_set_inf(res);
+    for (int w = MSM_NWIN - 1; w >= 0; w--) {
+        if (w != MSM_NWIN - 1) {
+            for (int d = 0; d < MSM_WBITS; d++) {
+                g1_jac t = res;
+                g1j_dbl(res, t);
+            }
+        }
+        g1_jac t = res;
+        g1j_add(res, t, wins[w]);
+    }
+    g1_affine a;
+    g1j_to_affine(a, res);
+    ff_to_bytes(out, a.x);
+    ff_to_bytes(out + 32, a.y);
+}
+
+int spectre_gpu_msm_g1_combine(const uint8_t* partials, uint32_t nshards,
+                               uint8_t out_affine[64]) {
+    if (!partials || nshards == 0) {
+        set_err("combine: bad args");
+        return -1;
+    }
+    g1_jac tot[MSM_NWIN];
+    for (int w = 0; w < MSM_NWIN; w++) g1j_set_inf(tot[w]);
+    for (uint32_t s = 0; s < nshards; s++) {  // deterministic rank order
+        const g1_jac* sh = (const g1_jac*)(partials + (size_t)s * MSM_NWIN * 96);
+        for (int w = 0; w < MSM_NWIN; w++) {
+            g1_jac t = tot[w];
+            g1j_add(tot[w], t, sh[w]);
+        }
+    }
+    winsums_to_affine(tot, out_affine);
+    return 0;
+}
+
+// ---------------------------------------------------------------- MSM
+int spectre_gpu_msm_g1_shard_device(spectre_gpu_ctx* ctx, int dev,
+                                    const void* d_bases, const void* d_scalars,
+                                    uint64_t n, uint32_t flags,
+                                    uint8_t* out_partials) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    return msm_shard_device(ctx, dev, (const g1_affine*)d_bases,
+                            (const uint8_t*)d_scalars, n, flags,
+                            (g1_jac*)out_partials);
+}
+
+int spectre_gpu_msm_g1_shard_device_timed(spectre_gpu_ctx* ctx, int dev,
+                                          const void* d_bases,
+                                          const void* d_scalars, uint64_t n,
+                                          uint32_t flags,
+                                          uint8_t* out_partials,
+                                          double out_ms[8]) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    return msm_shard_device(ctx, dev, (const g1_affine*)d_bases,
+                            (const uint8_t*)d_scalars, n, flags,
+                            (g1_jac*)out_partials, out_ms);
+}
+
+int spectre_gpu_msm_g1_device(spectre_gpu_ctx* ctx, int dev,
+                              const void* d_bases, const void* d_scalars,
+                              uint64_t n, uint32_t flags,
+                              uint8_t out_affine[64]) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    g1_jac wins[MSM_NWIN];
+    int rc = msm_shard_device(ctx, dev, (const g1_affine*)d_bases,
+                              (const uint8_t*)d_scalars, n, flags, wins);
+    if (rc) return rc;
+    winsums_to_affine(wins, out_affine);
+    return 0;
+}
+
+int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
+                       const uint8_t* bases, const uint8_t* scalars, uint64_t n,
+                       uint32_t flags, int num_gpus, uint8_t out_affine[64]) {
+    if (!ctx) {
+        set_err("null ctx");
+        return -1;
+    }
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    if (n == 0) {
+        memset(out_affine, 0, 64);
+        return 0;
+    }
+    if (num_gpus <= 0) num_gpus = 1;
+    if (num_gpus > (int)ctx->devs.size()) {
+        set_err("num_gpus %d > ctx devices %zu", num_gpus, ctx->devs.size());
+        return -1;
+    }
+    if (!scalars) {
+        set_err("scalars is NULL");
+        return -1;
+    }
+    std::vector<g1_jac> wins((size_t)num_gpus * MSM_NWIN);
+    for (int d = 0; d < num_gpus; d++) {
+        DeviceState& ds = ctx->devs[d];
+        HIP_TRY(hipSetDevice(ds.device_id));
+        const uint64_t lo = n * d / num_gpus, hi = n * (d + 1) / num_gpus;
+        const uint64_t m = hi - lo;
+        // scalars: plain scratch upload
+        if (ds.scal_cap < m * 32) {
+            if (ds.d_scalars) (void)hipFree(ds.d_scalars);
+            HIP_TRY(hipMalloc(&ds.d_scalars, m * 32));
+            ds.scal_cap = m * 32;
+        }
+        HIP_TRY(hipMemcpyAsync(ds.d_scalars, scalars + lo * 32, m * 32,
+                               hipMemcpyHostToDevice, ds.stream));
+        // bases: cached per (bases_id, n, num_gpus) — shard layout depends on
+        // all three, so they are all folded into the cache key.
+        g1_affine* d_b = nullptr;
+        if (bases_id != 0) {
+            auto key = std::make_pair(bases_id, n * 16 + (uint64_t)num_gpus);
+            auto it = ds.bases_cache.find(key);
+            if (it != ds.bases_cache.end()) {
+                d_b = it->second.d_ptr;
+            } else {
+                if (!bases) {
+                    set_err("bases_id %llu not cached and bases is NULL",
+                            (unsigned long long)bases_id);
+                    return -4;
+                }
+                CachedBases cb;
+                cb.n = m;
+                HIP_TRY(hipMalloc(&cb.d_ptr, m * sizeof(g1_affine)));
+                HIP_TRY(hipMemcpyAsync(cb.d_ptr, bases + lo * 64, m * 64,
+                                       hipMemcpyHostToDevice, ds.stream));
+                ds.bases_cache.emplace(key, cb);
+                d_b = cb.d_ptr;
+            }
+        } else {
+            if (!bases) {
+                set_err("bases is NULL");
+                return -1;
+            }
+            if (ds.base_cap < m) {
+                if (ds.d_bases) (void)hipFree(ds.d_bases);
+                HIP_TRY(hipMalloc(&ds.d_bases, m * sizeof(g1_affine)));
+                ds.base_cap = m;
+            }
+            HIP_TRY(hipMemcpyAsync(ds.d_bases, bases + lo * 64, m * 64,
+                                   hipMemcpyHostToDevice, ds.stream));
+            d_b = ds.d_bases;
+        }
+        int rc = msm_shard_device(ctx, d, d_b, ds.d_scalars, m, flags,
+                                  &wins[(size_t)d * MSM_NWIN]);
+        if (rc) return rc;
+    }
+    return spectre_gpu_msm_g1_combine((const uint8_t*)wins.data(),
+                                      (uint32_t)num_gpus, out_affine);
+}
+
+// ---------------------------------------------------------------- NTT
+int spectre_gpu_ntt_fr_device(spectre_gpu_ctx* ctx, int dev, void* d_data,
+                              uint32_t log_n, const uint8_t omega[32],
+                              int inverse, const uint8_t* coset_gen) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    fp256 om, g;
+    ff_from_bytes(om, omega);
+    if (coset_gen) ff_from_bytes(g, coset_gen);
+    return ntt_device(ctx, dev, (fp256*)d_data, log_n, om, inverse,
+                      coset_gen ? &g : nullptr);
+}
+
+int spectre_gpu_ntt_fr(spectre_gpu_ctx* ctx, uint8_t* data, uint32_t log_n,
+                       const uint8_t omega[32], int inverse,
+                       const uint8_t* coset_gen) {
+    if (check_dev(ctx, 0)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    DeviceState& ds = ctx->devs[0];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    const uint64_t n = 1ull << log_n;
+    void* d_buf = nullptr;
+    HIP_TRY(hipMalloc(&d_buf, n * 32));
+    HIP_TRY(hipMemcpyAsync(d_buf, data, n * 32, hipMemcpyHostToDevice,
+                           ds.stream));
+    int rc = spectre_gpu_ntt_fr_device(ctx, 0, d_buf, log_n, omega, inverse,
+                                       coset_gen);
+    if (rc == 0) {
+        rc = (hipMemcpy(data, d_buf, n * 32, hipMemcpyDeviceToHost) ==
+              hipSuccess)
+                 ? 0
+                 : -2;
+    }
+    (void)hipFree(d_buf);
+    return rc;
+}
+
+// ---------------------------------------------------------------- memory
+int spectre_gpu_malloc(spectre_gpu_ctx* ctx, int dev, size_t bytes,
+                       void** d_ptr) {
+    if (check_dev(ctx, dev)) return -1;
+    HIP_TRY(hipSetDevice(ctx->devs[dev].device_id));
+    HIP_TRY(hipMalloc(d_ptr, bytes));
+    return 0;
+}
+int spectre_gpu_free(spectre_gpu_ctx* ctx, int dev, void* d_ptr) {
+    if (check_dev(ctx, dev)) return -1;
+    HIP_TRY(hipSetDevice(ctx->devs[dev].device_id));
+    HIP_TRY(hipFree(d_ptr));
+    return 0;
+}
+int spectre_gpu_upload(spectre_gpu_ctx* ctx, int dev, void* d_dst,
+                       const void* src, size_t bytes) {
+    if (check_dev(ctx, dev)) return -1;
+    HIP_TRY(hipSetDevice(ctx->devs[dev].device_id));
+    HIP_TRY(hipMemcpy(d_dst, src, bytes, hipMemcpyHostToDevice));
+    return 0;
+}
+int spectre_gpu_download(spectre_gpu_ctx* ctx, int dev, void* dst,
+                         const void* d_src, size_t bytes) {
+    if (check_dev(ctx, dev)) return -1;
+    HIP_TRY(hipSetDevice(ctx->devs[dev].device_id));
+    HIP_TRY(hipMemcpy(dst, d_src, bytes, hipMemcpyDeviceToHost));
+    return 0;
+}
+int spectre_gpu_synchronize(spectre_gpu_ctx* ctx, int dev) {
+    if (check_dev(ctx, dev)) return -1;
+    HIP_TRY(hipSetDevice(ctx->devs[dev].device_id));
+    HIP_TRY(hipDeviceSynchronize());
+    return 0;
+}
+
+}  // extern "C"

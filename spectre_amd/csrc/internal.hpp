@@ -1,0 +1,94 @@
+// internal.hpp — shared internals of libspectre_gpu.so (not installed).
+#pragma once
+#include <hip/hip_runtime.h>
+#include "spectre_gpu.h"
+#include "ff.hpp"
+#include "g1.hpp"
+#include <array>
+#include <map>
+#include <mutex>
+#include <string>
+#include <vector>
+
+// MSM window configuration: 16 signed 16-bit windows over 256-bit scalars.
+// Signed-digit recoding halves the bucket count: digit magnitudes in
+// [1, 2^15], bucket index = magnitude-1, negative digits negate the point
+// (affine negation is one Fq negation). Canonical BN254 Fr scalars are
+// < 2^254, so window 15 never overflows and never carries out.
+#define MSM_NWIN 16
+#define MSM_WBITS 16
+#define MSM_BPW 32768u                     // buckets per window
+#define MSM_NB_TOTAL (MSM_NWIN * MSM_BPW)  // 524288
+#define MSM_SKIP_KEY MSM_NB_TOTAL          // sorts after all real keys
+#define MSM_SORT_BITS 20                   // key range < 2^20
+#define MSM_CHUNK 16                       // buckets per reduction thread
+
+struct NttPlan {
+    fp256* tw1 = nullptr;  // (w^{n2})^j, j < n1/2
+    fp256* tw2 = nullptr;  // (w^{n1})^j, j < n2/2
+    fp256* twB = nullptr;  // w^c, c < n2 (inter-pass twiddle bases)
+    uint32_t k1 = 0, k2 = 0;
+};
+
+struct CachedBases {
+    g1_affine* d_ptr = nullptr;
+    uint64_t n = 0;
+};
+
+struct DeviceState {
+    int device_id = 0;
+    hipStream_t stream = nullptr;
+    // ---- MSM scratch (grown on demand) ----
+    uint32_t* d_keys_in = nullptr;
+    uint32_t* d_keys_out = nullptr;
+    uint32_t* d_vals_in = nullptr;
+    uint32_t* d_vals_out = nullptr;
+    size_t ent_cap = 0;  // capacity in entries (16*n)
+    void* d_sort_tmp = nullptr;
+    size_t sort_tmp_cap = 0;
+    uint32_t* d_offsets = nullptr;  // MSM_NB_TOTAL + 1
+    g1_jac* d_buckets = nullptr;    // MSM_NB_TOTAL
+    g1_jac* d_red = nullptr;        // reduction ping-pong (NB_TOTAL/CHUNK * 2)
+    uint8_t* d_scalars = nullptr;
+    size_t scal_cap = 0;  // bytes
+    g1_affine* d_bases = nullptr;
+    size_t base_cap = 0;  // points
+    std::map<std::pair<uint64_t, uint64_t>, CachedBases> bases_cache;  // (id, n)
+    // ---- NTT scratch ----
+    fp256* d_ntt_tmp = nullptr;
+    size_t ntt_cap = 0;  // elements
+    fp256* d_cosetA = nullptr;  // coset base tables (n2 resp. n1 entries)
+    fp256* d_cosetB = nullptr;
+    size_t coset_cap = 0;
+    std::map<std::array<uint8_t, 40>, NttPlan> plans;  // omega||log_n||inverse
+};
+
+struct spectre_gpu_ctx {
+    std::vector<DeviceState> devs;
+    std::recursive_mutex mu;
+};
+
+void set_err(const char* fmt, ...);
+
+#define HIP_TRY(x)                                                        \
+    do {                                                                  \
+        hipError_t _e = (x);                                              \
+        if (_e != hipSuccess) {                                           \
+            set_err("%s:%d %s: %s", __FILE__, __LINE__, #x,               \
+                    hipGetErrorString(_e));                               \
+            return -2;                                                    \
+        }                                                                 \
+    } while (0)
+
+// msm.hip — runs the full Pippenger pipeline for one shard on device `dev`'s
+// stream and writes the MSM_NWIN Jacobian window sums to host memory
+// (synchronizes the stream). If stage_ms != nullptr, per-stage HIP-event
+// timings are written: [0]=digits [1]=sort [2]=offsets [3]=bucket_acc
+// [4]=chunks [5]=reduce [6]=total-gpu [7]=real (non-zero-digit) entry count.
+int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
+                     const uint8_t* d_scalars, uint64_t n, uint32_t flags,
+                     g1_jac* winsums_host, double* stage_ms = nullptr);
+
+// ntt.hip — in-place NTT on a device buffer (synchronizes the stream).
+int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
+               const fp256& omega, int inverse, const fp256* coset_gen);

@@ -1,0 +1,281 @@
+// msm.hip — BN254 G1 multi-scalar multiplication (Pippenger) for gfx950.
+//
+// Computes what halo2curves-axiom 0.5.2 `best_multiexp` computes (the value
+// Sum_i k_i * P_i; windowing is free — see oracle/bn254.c header for the
+// parity contract). MI355X-native structure:
+//   1. k_msm_digits     — one thread per scalar: optional Montgomery->canonical
+//                         conversion in registers, signed 16-bit window
+//                         recoding, emit (bucket key, point index|sign) pairs.
+//                         Writes are window-major (w*n + i) so each window's
+//                         entries are written coalesced.
+//   2. radix sort       — hipCUB DeviceRadixSort on the 20-bit keys
+//                         (16 windows x 2^15 buckets; zero digits get a
+//                         sentinel key that sorts last and is never touched).
+//   3. k_bucket_offsets — one thread per bucket: binary-search the sorted key
+//                         stream for the bucket's segment bounds.
+//   4. k_bucket_acc     — one thread per bucket: serial Jacobian+affine mixed
+//                         adds over its segment. Deterministic by
+//                         construction (segment order is index order after
+//                         the stable sort); the affine result is canonical,
+//                         so ANY schedule yields bit-identical output bytes.
+//   5. k_window_chunks  — one thread per MSM_CHUNK consecutive buckets:
+//                         weighted suffix sum within the chunk, then one
+//                         15-bit double-and-add to lift the chunk to its
+//                         window offset. Independent across threads (no
+//                         serial suffix chain across the whole window).
+//   6. k_reduce_groups  — log-tree group sums down to MSM_NWIN window sums.
+// Host side (ffi.cpp) finishes with 15 Horner steps (16 doubl-groups + adds)
+// and one field inversion to affine — microseconds of host work.
+//
+// All field math is 8x32-limb Montgomery CIOS (ff.hpp) — VALU integer work;
+// MFMA does not apply to modular arithmetic.
+#include "internal.hpp"
+#include <hipcub/hipcub.hpp>
+
+#define THREADS 256
+
+// ---- kernel 1: signed window decomposition --------------------------------
+__global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
+                             int canonical, uint32_t* __restrict__ keys,
+                             uint32_t* __restrict__ vals) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    fp256 s;
+    ff_from_bytes(s, scalars + 32 * i);
+    if (!canonical) ff_from_mont<Fr>(s, s);
+    uint32_t carry = 0;
+    for (int w = 0; w < MSM_NWIN; w++) {
+        // 16-bit window w of the canonical scalar
+        uint32_t limb = s.l[w >> 1];
+        uint32_t d = ((w & 1) ? (limb >> 16) : (limb & 0xffffu)) + carry;
+        uint32_t key, val = (uint32_t)i;
+        if (d == 0) {
+            carry = 0;
+            key = MSM_SKIP_KEY;
+        } else if (d <= 32768u) {  // positive digit, magnitude d (2^15 kept +)
+            carry = 0;
+            key = w * MSM_BPW + (d - 1);
+        } else if (d == 65536u) {  // 0xffff + carry: digit 0, carry out
+            carry = 1;
+            key = MSM_SKIP_KEY;
+        } else {  // negative digit, magnitude 2^16 - d
+            carry = 1;
+            key = w * MSM_BPW + (65536u - d - 1);
+            val |= 0x80000000u;
+        }
+        keys[(uint64_t)w * n + i] = key;
+        vals[(uint64_t)w * n + i] = val;
+    }
+}
+
+// ---- kernel 3: per-bucket segment bounds via binary search ----------------
+__global__ void k_bucket_offsets(const uint32_t* __restrict__ keys,
+                                 uint64_t nent, uint32_t* __restrict__ off) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b > MSM_NB_TOTAL) return;
+    // lower_bound(keys, b)
+    uint64_t lo = 0, hi = nent;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1;
+        if (keys[mid] < b) lo = mid + 1;
+        else hi = mid;
+    }
+    off[b] = (uint32_t)lo;
+}
+
+// ---- kernel 4: bucket accumulation (mixed adds) ---------------------------
+__global__ void k_bucket_acc(const uint32_t* __restrict__ off,
+                             const uint32_t* __restrict__ vals,
+                             const g1_affine* __restrict__ bases,
+                             g1_jac* __restrict__ buckets) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= MSM_NB_TOTAL) return;
+    uint32_t s = off[b], e = off[b + 1];
+    g1_jac acc;
+    g1j_set_inf(acc);
+    for (uint32_t j = s; j < e; j++) {
+        uint32_t v = vals[j];
+        g1_affine p = bases[v & 0x7fffffffu];
+        if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
+        g1_jac t = acc;
+        g1j_add_affine(acc, t, p);
+    }
+    buckets[b] = acc;
+}
+
+// ---- kernel 5: weighted chunk reduction -----------------------------------
+// For bucket-local index j in a window, the bucket's multiplier is (j+1).
+// Chunk covering window-local buckets [m, m+CHUNK):
+//   sum_{j=0}^{CHUNK-1} (j+1) * B[m+j]  +  m * sum_j B[m+j]
+// first term by suffix running sums, second by double-and-add (m < 2^15).
+__global__ void k_window_chunks(const g1_jac* __restrict__ buckets,
+                                g1_jac* __restrict__ out) {
+    const uint32_t nchunks_pw = MSM_BPW / MSM_CHUNK;
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= MSM_NB_TOTAL / MSM_CHUNK) return;
+    uint32_t w = t / nchunks_pw;
+    uint32_t cw = t % nchunks_pw;
+    const g1_jac* B = buckets + (uint64_t)w * MSM_BPW + (uint64_t)cw * MSM_CHUNK;
+    g1_jac accT, accW;
+    g1j_set_inf(accT);
+    g1j_set_inf(accW);
+    for (int j = MSM_CHUNK - 1; j >= 0; j--) {
+        g1_jac x = accT;
+        g1j_add(accT, x, B[j]);
+        x = accW;
+        g1j_add(accW, x, accT);
+    }
+    // accW += (cw*CHUNK) * accT
+    uint32_t m = cw * MSM_CHUNK;
+    if (m && !g1j_is_inf(accT)) {
+        g1_jac acc;
+        g1j_set_inf(acc);
+        for (int bit = 15; bit >= 0; bit--) {
+            g1_jac x = acc;
+            g1j_dbl(acc, x);
+            if ((m >> bit) & 1) {
+                x = acc;
+                g1j_add(acc, x, accT);
+            }
+        }
+        g1_jac x = accW;
+        g1j_add(accW, x, acc);
+    }
+    out[t] = accW;
+}
+
+// ---- kernel 6: grouped tree sum -------------------------------------------
+__global__ void k_reduce_groups(const g1_jac* __restrict__ in,
+                                g1_jac* __restrict__ out, uint32_t group,
+                                uint32_t total_out) {
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= total_out) return;
+    g1_jac acc;
+    g1j_set_inf(acc);
+    for (uint32_t j = 0; j < group; j++) {
+        g1_jac x = acc;
+        g1j_add(acc, x, in[(uint64_t)t * group + j]);
+    }
+    out[t] = acc;
+}
+
+// ---- host orchestration ---------------------------------------------------
+static int ensure_msm_scratch(DeviceState& ds, uint64_t n) {
+    const uint64_t ent = (uint64_t)MSM_NWIN * n;
+    if (ds.ent_cap < ent) {
+        for (void* p : {(void*)ds.d_keys_in, (void*)ds.d_keys_out,
+                        (void*)ds.d_vals_in, (void*)ds.d_vals_out})
+            if (p) (void)hipFree(p);
+        HIP_TRY(hipMalloc(&ds.d_keys_in, ent * 4));
+        HIP_TRY(hipMalloc(&ds.d_keys_out, ent * 4));
+        HIP_TRY(hipMalloc(&ds.d_vals_in, ent * 4));
+        HIP_TRY(hipMalloc(&ds.d_vals_out, ent * 4));
+        ds.ent_cap = ent;
+    }
+    if (!ds.d_offsets) HIP_TRY(hipMalloc(&ds.d_offsets, (MSM_NB_TOTAL + 1) * 4));
+    if (!ds.d_buckets)
+        HIP_TRY(hipMalloc(&ds.d_buckets, (uint64_t)MSM_NB_TOTAL * sizeof(g1_jac)));
+    if (!ds.d_red)
+        HIP_TRY(hipMalloc(&ds.d_red,
+                          (uint64_t)(MSM_NB_TOTAL / MSM_CHUNK) * 2 * sizeof(g1_jac)));
+    size_t sort_need = 0;
+    (void)hipcub::DeviceRadixSort::SortPairs(nullptr, sort_need, ds.d_keys_in,
+                                       ds.d_keys_out, ds.d_vals_in,
+                                       ds.d_vals_out, (int64_t)ent, 0,
+                                       MSM_SORT_BITS, ds.stream);
+    if (ds.sort_tmp_cap < sort_need) {
+        if (ds.d_sort_tmp) (void)hipFree(ds.d_sort_tmp);
+        HIP_TRY(hipMalloc(&ds.d_sort_tmp, sort_need));
+        ds.sort_tmp_cap = sort_need;
+    }
+    return 0;
+}
+
+int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
+                     const uint8_t* d_scalars, uint64_t n, uint32_t flags,
+                     g1_jac* winsums_host, double* stage_ms) {
+    DeviceState& ds = ctx->devs[dev];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    if (n == 0) {
+        for (int w = 0; w < MSM_NWIN; w++) g1j_set_inf(winsums_host[w]);
+        return 0;
+    }
+    int rc = ensure_msm_scratch(ds, n);
+    if (rc) return rc;
+    const uint64_t ent = (uint64_t)MSM_NWIN * n;
+    const int canonical = (flags & SPECTRE_SCALARS_CANONICAL) ? 1 : 0;
+    hipStream_t st = ds.stream;
+
+    // optional per-stage HIP events (on the library stream — this is the
+    // live in-bench measurement the roofline report is built from)
+    hipEvent_t ev[7];
+    if (stage_ms)
+        for (auto& e : ev) HIP_TRY(hipEventCreate(&e));
+#define STAMP(i)                                  \
+    if (stage_ms) HIP_TRY(hipEventRecord(ev[i], st));
+
+    STAMP(0);
+    uint32_t blocks = (uint32_t)((n + THREADS - 1) / THREADS);
+    hipLaunchKernelGGL(k_msm_digits, dim3(blocks), dim3(THREADS), 0, st,
+                       d_scalars, n, canonical, ds.d_keys_in, ds.d_vals_in);
+    STAMP(1);
+    size_t tmp = ds.sort_tmp_cap;
+    (void)hipcub::DeviceRadixSort::SortPairs(ds.d_sort_tmp, tmp, ds.d_keys_in,
+                                       ds.d_keys_out, ds.d_vals_in,
+                                       ds.d_vals_out, (int64_t)ent, 0,
+                                       MSM_SORT_BITS, st);
+    STAMP(2);
+    hipLaunchKernelGGL(k_bucket_offsets,
+                       dim3((MSM_NB_TOTAL + 1 + THREADS - 1) / THREADS),
+                       dim3(THREADS), 0, st, ds.d_keys_out, ent, ds.d_offsets);
+    STAMP(3);
+    hipLaunchKernelGGL(k_bucket_acc, dim3(MSM_NB_TOTAL / THREADS),
+                       dim3(THREADS), 0, st, ds.d_offsets, ds.d_vals_out,
+                       d_bases, ds.d_buckets);
+    STAMP(4);
+    const uint32_t nchunks = MSM_NB_TOTAL / MSM_CHUNK;  // 32768
+    g1_jac* red0 = ds.d_red;
+    g1_jac* red1 = ds.d_red + nchunks;
+    hipLaunchKernelGGL(k_window_chunks, dim3(nchunks / THREADS), dim3(THREADS),
+                       0, st, ds.d_buckets, red0);
+    STAMP(5);
+    // tree: 32768 -> 2048 -> 128 -> 16 (groups of 16, 16, 8)
+    uint32_t cur = nchunks;
+    const uint32_t groups[3] = {16, 16, 8};
+    g1_jac* in = red0;
+    g1_jac* out = red1;
+    for (int lvl = 0; lvl < 3; lvl++) {
+        uint32_t g = groups[lvl];
+        uint32_t tot = cur / g;
+        hipLaunchKernelGGL(k_reduce_groups,
+                           dim3((tot + THREADS - 1) / THREADS), dim3(THREADS),
+                           0, st, in, out, g, tot);
+        g1_jac* t = in;
+        in = out;
+        out = t;
+        cur = tot;
+    }
+    STAMP(6);
+    // `in` now holds MSM_NWIN window sums
+    HIP_TRY(hipMemcpyAsync(winsums_host, in, MSM_NWIN * sizeof(g1_jac),
+                           hipMemcpyDeviceToHost, st));
+    uint32_t ent_real = 0;
+    if (stage_ms)
+        HIP_TRY(hipMemcpyAsync(&ent_real, ds.d_offsets + MSM_NB_TOTAL, 4,
+                               hipMemcpyDeviceToHost, st));
+    HIP_TRY(hipStreamSynchronize(st));
+    HIP_TRY(hipGetLastError());
+    if (stage_ms) {
+        float ms;
+        for (int i = 0; i < 6; i++) {
+            HIP_TRY(hipEventElapsedTime(&ms, ev[i], ev[i + 1]));
+            stage_ms[i] = ms;
+        }
+        HIP_TRY(hipEventElapsedTime(&ms, ev[0], ev[6]));
+        stage_ms[6] = ms;
+        stage_ms[7] = (double)ent_real;
+        for (auto& e : ev) (void)hipEventDestroy(e);
+    }
+#undef STAMP
+    return 0;
+}

@@ -1,0 +1,259 @@
+// ntt.hip — radix-2 NTT over BN254 Fr for gfx950 (MI355X).
+//
+// Computes what halo2's best_fft / EvaluationDomain compute (the plain DFT
+// out[j] = Sum_i a[i] omega^(ij); inverse = DFT with omega^{-1} then * n^{-1};
+// coset = multiply by coset_gen^i before (forward) / after (inverse)).
+//
+// MI355X-native structure (four-step / Bailey decomposition, n = n1*n2,
+// n2 = min(2^12, n), so each sub-transform fits LDS and the whole transform
+// is TWO global passes instead of log2(n) — the difference between ~1 GiB
+// and ~11.5 GiB of HBM traffic at n=2^23):
+//   pass A (k_ntt_col, only when n1 > 1): one workgroup per column i2:
+//     column DFT over i1 (stride n2), in-LDS DIF (bit-reversed output order
+//     is folded into the store index, costing nothing extra — the column
+//     store is strided anyway), fused inter-pass twiddle omega^(i2*t1) and
+//     optional forward-coset pre-multiply.
+//   pass B (k_ntt_row): one workgroup per row t1: contiguous (coalesced)
+//     row DFT over i2, in-LDS DIF, store transposed to out[t2*n1 + t1] with
+//     fused n^{-1} scaling and optional inverse-coset post-multiply.
+// Twiddle/coset powers g^e (e < n <= 2^24) resolve via two 4096-entry tables:
+//   g^e = T1[e & 0xfff] * T2[e >> 12]   (one mul + two cached loads)
+// built once per (omega, log_n) plan and cached on device.
+//
+// All Fr math is 8x32-limb Montgomery (ff.hpp); data stays in Montgomery form
+// end-to-end exactly as halo2 holds its &[Fr] slices.
+#include "internal.hpp"
+
+#define THREADS 256
+#define TW_LOW_BITS 12
+#define TW_LOW_MASK 0xfffu
+
+__device__ __forceinline__ uint32_t bitrev(uint32_t x, uint32_t bits) {
+    return bits ? (__brev(x) >> (32 - bits)) : 0;
+}
+// g^e via the 2D power table (e < n <= 2^24)
+__device__ __forceinline__ void tw_lookup(fp256& o, const fp256* T1,
+                                          const fp256* T2, uint32_t e) {
+    ff_mul<Fr>(o, T1[e & TW_LOW_MASK], T2[e >> TW_LOW_BITS]);
+}
+
+// out[j] = base^j, j < count
+__global__ void k_pow_table(fp256 base, fp256* __restrict__ out,
+                            uint32_t count) {
+    uint32_t j = blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= count) return;
+    fp256 v;
+    ff_pow_u32<Fr>(v, base, j);
+    out[j] = v;
+}
+
+// in-LDS DIF butterflies over L = 2^logL elements; twL[j] = (root)^j, j < L/2.
+// On exit lds[s] holds DFT output index bitrev(s, logL).
+__device__ void lds_dif(fp256* lds, const fp256* __restrict__ twL,
+                        uint32_t logL) {
+    const uint32_t L = 1u << logL;
+    for (uint32_t h = L >> 1; h >= 1; h >>= 1) {
+        const uint32_t stride = (L >> 1) / h;
+        for (uint32_t p = threadIdx.x; p < (L >> 1); p += blockDim.x) {
+            uint32_t blk = p / h, j = p % h;
+            uint32_t i0 = blk * 2 * h + j, i1 = i0 + h;
+            fp256 u = lds[i0], v = lds[i1], t;
+            ff_add<Fr>(lds[i0], u, v);
+            ff_sub<Fr>(t, u, v);
+            ff_mul<Fr>(lds[i1], t, twL[(uint64_t)j * stride]);
+        }
+        __syncthreads();
+    }
+}
+
+// pass A: column DFTs. grid.x = n2; LDS = n1 elements.
+__global__ void k_ntt_col(const fp256* __restrict__ in, fp256* __restrict__ out,
+                          const fp256* __restrict__ tw1,
+                          const fp256* __restrict__ T1,
+                          const fp256* __restrict__ T2,
+                          const fp256* __restrict__ cT1,
+                          const fp256* __restrict__ cT2, uint32_t log_n1,
+                          uint32_t log_n2) {
+    extern __shared__ fp256 lds[];
+    const uint32_t n1 = 1u << log_n1;
+    const uint32_t n2 = 1u << log_n2;
+    const uint32_t c = blockIdx.x;
+    for (uint32_t s = threadIdx.x; s < n1; s += blockDim.x) {
+        fp256 v = in[(uint64_t)s * n2 + c];
+        if (cT1) {  // forward coset: multiply by g^(global index)
+            fp256 f;
+            tw_lookup(f, cT1, cT2, s * n2 + c);
+            ff_mul<Fr>(v, v, f);
+        }
+        lds[s] = v;
+    }
+    __syncthreads();
+    lds_dif(lds, tw1, log_n1);
+    // store with inter-pass twiddle omega^(c * t1), t1 = bitrev(s)
+    for (uint32_t s = threadIdx.x; s < n1; s += blockDim.x) {
+        uint32_t t1 = bitrev(s, log_n1);
+        fp256 f, v;
+        tw_lookup(f, T1, T2, c * t1);  // c*t1 < n2*n1 = n
+        ff_mul<Fr>(v, lds[s], f);
+        out[(uint64_t)t1 * n2 + c] = v;
+    }
+}
+
+// pass B: row DFTs + transposed store. grid.x = n1; LDS = n2 elements.
+// in == out is safe only when n1 == 1 (single workgroup).
+__global__ void k_ntt_row(const fp256* __restrict__ in, fp256* __restrict__ out,
+                          const fp256* __restrict__ tw2,
+                          const fp256* __restrict__ cT1,
+                          const fp256* __restrict__ cT2, int coset_on_load,
+                          fp256 scale, int apply_scale, uint32_t log_n1,
+                          uint32_t log_n2) {
+    extern __shared__ fp256 lds[];
+    const uint32_t n1 = 1u << log_n1;
+    const uint32_t n2 = 1u << log_n2;
+    const uint32_t r = blockIdx.x;
+    for (uint32_t s = threadIdx.x; s < n2; s += blockDim.x) {
+        fp256 v = in[(uint64_t)r * n2 + s];
+        if (cT1 && coset_on_load) {  // 1-pass forward coset (r == 0)
+            fp256 f;
+            tw_lookup(f, cT1, cT2, s);
+            ff_mul<Fr>(v, v, f);
+        }
+        lds[s] = v;
+    }
+    __syncthreads();
+    lds_dif(lds, tw2, log_n2);
+    for (uint32_t s = threadIdx.x; s < n2; s += blockDim.x) {
+        uint32_t t2 = bitrev(s, log_n2);
+        fp256 v = lds[s];
+        if (apply_scale) ff_mul<Fr>(v, v, scale);
+        if (cT1 && !coset_on_load) {  // inverse coset: g^(output index)
+            fp256 f;
+            tw_lookup(f, cT1, cT2, t2 * n1 + r);
+            ff_mul<Fr>(v, v, f);
+        }
+        out[(uint64_t)t2 * n1 + r] = v;
+    }
+}
+
+// ---------------------------------------------------------------- host side
+static void host_pow_u32(fp256& o, const fp256& a, uint32_t e) {
+    ff_pow_u32<Fr>(o, a, e);
+}
+
+static int build_tables(DeviceState& ds, const fp256& base, fp256* out,
+                        uint32_t count) {
+    hipLaunchKernelGGL(k_pow_table, dim3((count + THREADS - 1) / THREADS),
+                       dim3(THREADS), 0, ds.stream, base, out, count);
+    return 0;
+}
+
+// build-or-get the cached twiddle plan for (omega, log_n)
+static int get_plan(DeviceState& ds, const fp256& omega, uint32_t log_n,
+                    NttPlan** out) {
+    std::array<uint8_t, 40> key{};
+    memcpy(key.data(), omega.l, 32);
+    memcpy(key.data() + 32, &log_n, 4);
+    auto it = ds.plans.find(key);
+    if (it != ds.plans.end()) {
+        *out = &it->second;
+        return 0;
+    }
+    NttPlan p;
+    p.k2 = log_n < 12 ? log_n : 12;
+    p.k1 = log_n - p.k2;
+    const uint32_t n1 = 1u << p.k1, n2 = 1u << p.k2;
+    const uint32_t t1n = (log_n < TW_LOW_BITS) ? (1u << log_n) : (1u << TW_LOW_BITS);
+    const uint32_t t2n = (log_n > TW_LOW_BITS) ? (1u << (log_n - TW_LOW_BITS)) : 1;
+    if (p.k1) HIP_TRY(hipMalloc(&p.tw1, (n1 / 2) * sizeof(fp256)));
+    HIP_TRY(hipMalloc(&p.tw2, (n2 / 2 ? n2 / 2 : 1) * sizeof(fp256)));
+    HIP_TRY(hipMalloc(&p.twB, (t1n + t2n) * sizeof(fp256)));  // T1 || T2
+    fp256 w1, w2, wT2;
+    host_pow_u32(w1, omega, n2);                    // root of column DFT
+    host_pow_u32(w2, omega, n1);                    // root of row DFT
+    host_pow_u32(wT2, omega, 1u << TW_LOW_BITS);    // T2 base
+    if (p.k1) build_tables(ds, w1, p.tw1, n1 / 2);
+    build_tables(ds, w2, p.tw2, n2 / 2 ? n2 / 2 : 1);
+    build_tables(ds, omega, p.twB, t1n);
+    build_tables(ds, wT2, p.twB + t1n, t2n);
+    HIP_TRY(hipStreamSynchronize(ds.stream));
+    HIP_TRY(hipGetLastError());
+    auto res = ds.plans.emplace(key, p);
+    *out = &res.first->second;
+    return 0;
+}
+
+int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
+               const fp256& omega, int inverse, const fp256* coset_gen) {
+    DeviceState& ds = ctx->devs[dev];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    if (log_n > 24) {
+        set_err("ntt: log_n %u > 24 unsupported", log_n);
+        return -3;
+    }
+    if (log_n == 0) return 0;  // DFT of size 1 is the identity; n^{-1} = 1, g^0 = 1
+    const uint64_t n = 1ull << log_n;
+    NttPlan* plan = nullptr;
+    int rc = get_plan(ds, omega, log_n, &plan);
+    if (rc) return rc;
+    const uint32_t n1 = 1u << plan->k1, n2 = 1u << plan->k2;
+    const uint32_t t1n = (log_n < TW_LOW_BITS) ? (uint32_t)n : (1u << TW_LOW_BITS);
+    const uint32_t t2n = (log_n > TW_LOW_BITS) ? (uint32_t)(n >> TW_LOW_BITS) : 1;
+    // scratch
+    if (ds.ntt_cap < n) {
+        if (ds.d_ntt_tmp) (void)hipFree(ds.d_ntt_tmp);
+        HIP_TRY(hipMalloc(&ds.d_ntt_tmp, n * sizeof(fp256)));
+        ds.ntt_cap = n;
+    }
+    // coset power tables (per call; tiny)
+    fp256* cT1 = nullptr;
+    fp256* cT2 = nullptr;
+    if (coset_gen) {
+        if (ds.coset_cap < (size_t)t1n + t2n) {
+            if (ds.d_cosetA) (void)hipFree(ds.d_cosetA);
+            HIP_TRY(hipMalloc(&ds.d_cosetA, ((size_t)t1n + t2n) * sizeof(fp256)));
+            ds.coset_cap = (size_t)t1n + t2n;
+        }
+        cT1 = ds.d_cosetA;
+        cT2 = ds.d_cosetA + t1n;
+        fp256 gT2;
+        host_pow_u32(gT2, *coset_gen, 1u << TW_LOW_BITS);
+        build_tables(ds, *coset_gen, cT1, t1n);
+        build_tables(ds, gT2, cT2, t2n);
+    }
+    // n^{-1} scale for inverse
+    fp256 scale;
+    ff_set_one<Fr>(scale);
+    if (inverse) {
+        fp256 ncanon, nm;
+        ff_set_zero(ncanon);
+        ncanon.l[log_n >> 5] = 1u << (log_n & 31);
+        ff_to_mont<Fr>(nm, ncanon);
+        ff_inv<Fr>(scale, nm);
+    }
+    hipStream_t st = ds.stream;
+    const fp256* fwd_cT1 = (coset_gen && !inverse) ? cT1 : nullptr;
+    const fp256* fwd_cT2 = (coset_gen && !inverse) ? cT2 : nullptr;
+    const fp256* inv_cT1 = (coset_gen && inverse) ? cT1 : nullptr;
+    const fp256* inv_cT2 = (coset_gen && inverse) ? cT2 : nullptr;
+    if (plan->k1 > 0) {
+        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(THREADS),
+                           n1 * sizeof(fp256), st, d_data, ds.d_ntt_tmp,
+                           plan->tw1, plan->twB, plan->twB + t1n, fwd_cT1,
+                           fwd_cT2, plan->k1, plan->k2);
+        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(THREADS),
+                           n2 * sizeof(fp256), st, ds.d_ntt_tmp, d_data,
+                           plan->tw2, inv_cT1, inv_cT2, /*coset_on_load=*/0,
+                           scale, inverse ? 1 : 0, plan->k1, plan->k2);
+    } else {
+        // single pass; forward coset applies on load, inverse on store
+        const fp256* cc1 = coset_gen ? cT1 : nullptr;
+        const fp256* cc2 = coset_gen ? cT2 : nullptr;
+        hipLaunchKernelGGL(k_ntt_row, dim3(1), dim3(THREADS),
+                           n2 * sizeof(fp256), st, d_data, d_data, plan->tw2,
+                           cc1, cc2, /*coset_on_load=*/inverse ? 0 : 1, scale,
+                           inverse ? 1 : 0, 0, plan->k2);
+    }
+    HIP_TRY(hipStreamSynchronize(st));
+    HIP_TRY(hipGetLastError());
+    return 0;
+}

@@ -1,0 +1,240 @@
+"""ctypes binding over libspectre_gpu.so (the product C ABI).
+
+Mirrors include/spectre_gpu.h one-to-one. Compute entry points require a
+visible AMD GPU; only `combine_partials` (host-side final reduction) and
+`version` work without one — exactly the contract of the C library.
+"""
+import ctypes
+import os
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB = os.path.join(_HERE, "libspectre_gpu.so")
+
+SCALARS_MONTGOMERY = 0
+SCALARS_CANONICAL = 1
+NUM_WINDOWS = 16
+PARTIALS_BYTES = NUM_WINDOWS * 96  # per-shard Jacobian window sums
+
+_lib = None
+
+
+def lib_path() -> str:
+    return _LIB
+
+
+def load_library() -> ctypes.CDLL:
+    """dlopen the product library and declare signatures. Raises if missing —
+    build with `make -C spectre_amd/csrc` (or __graft_entry__.build())."""
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(_LIB):
+        raise RuntimeError(
+            f"libspectre_gpu.so not found at {_LIB}; run __graft_entry__.build()"
+        )
+    lib = ctypes.CDLL(_LIB)
+    c = ctypes
+    lib.spectre_gpu_init.restype = c.c_void_p
+    lib.spectre_gpu_init.argtypes = [c.c_int, c.POINTER(c.c_int)]
+    lib.spectre_gpu_destroy.argtypes = [c.c_void_p]
+    lib.spectre_gpu_last_error.restype = c.c_char_p
+    lib.spectre_gpu_version.restype = c.c_char_p
+    lib.spectre_gpu_device_count.restype = c.c_int
+    lib.spectre_gpu_device_count.argtypes = [c.c_void_p]
+    lib.spectre_gpu_msm_g1.restype = c.c_int
+    lib.spectre_gpu_msm_g1.argtypes = [
+        c.c_void_p, c.c_uint64, c.c_void_p, c.c_void_p, c.c_uint64,
+        c.c_uint32, c.c_int, c.c_void_p,
+    ]
+    lib.spectre_gpu_msm_g1_device.restype = c.c_int
+    lib.spectre_gpu_msm_g1_device.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
+        c.c_void_p,
+    ]
+    lib.spectre_gpu_msm_g1_shard_device.restype = c.c_int
+    lib.spectre_gpu_msm_g1_shard_device.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
+        c.c_void_p,
+    ]
+    lib.spectre_gpu_msm_g1_shard_device_timed.restype = c.c_int
+    lib.spectre_gpu_msm_g1_shard_device_timed.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
+        c.c_void_p, c.POINTER(c.c_double),
+    ]
+    lib.spectre_gpu_msm_g1_combine.restype = c.c_int
+    lib.spectre_gpu_msm_g1_combine.argtypes = [c.c_void_p, c.c_uint32, c.c_void_p]
+    lib.spectre_gpu_ntt_fr.restype = c.c_int
+    lib.spectre_gpu_ntt_fr.argtypes = [
+        c.c_void_p, c.c_void_p, c.c_uint32, c.c_void_p, c.c_int, c.c_void_p,
+    ]
+    lib.spectre_gpu_ntt_fr_device.restype = c.c_int
+    lib.spectre_gpu_ntt_fr_device.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_uint32, c.c_void_p, c.c_int,
+        c.c_void_p,
+    ]
+    lib.spectre_gpu_malloc.restype = c.c_int
+    lib.spectre_gpu_malloc.argtypes = [c.c_void_p, c.c_int, c.c_size_t,
+                                       c.POINTER(c.c_void_p)]
+    lib.spectre_gpu_free.restype = c.c_int
+    lib.spectre_gpu_free.argtypes = [c.c_void_p, c.c_int, c.c_void_p]
+    lib.spectre_gpu_upload.restype = c.c_int
+    lib.spectre_gpu_upload.argtypes = [c.c_void_p, c.c_int, c.c_void_p,
+                                       c.c_void_p, c.c_size_t]
+    lib.spectre_gpu_download.restype = c.c_int
+    lib.spectre_gpu_download.argtypes = [c.c_void_p, c.c_int, c.c_void_p,
+                                         c.c_void_p, c.c_size_t]
+    lib.spectre_gpu_synchronize.restype = c.c_int
+    lib.spectre_gpu_synchronize.argtypes = [c.c_void_p, c.c_int]
+    _lib = lib
+    return lib
+
+
+def version() -> str:
+    return load_library().spectre_gpu_version().decode()
+
+
+def combine_partials(partials: bytes, nshards: int) -> bytes:
+    """Host-only: combine shard window partials (rank order) to affine."""
+    lib = load_library()
+    assert len(partials) == nshards * PARTIALS_BYTES
+    buf = (ctypes.c_uint8 * len(partials)).from_buffer_copy(partials)
+    out = (ctypes.c_uint8 * 64)()
+    rc = lib.spectre_gpu_msm_g1_combine(buf, nshards, out)
+    if rc != 0:
+        raise RuntimeError(f"combine failed rc={rc}: "
+                           f"{lib.spectre_gpu_last_error().decode()}")
+    return bytes(out)
+
+
+class SpectreGpu:
+    """A context over one or more MI355X devices. Raises without a GPU."""
+
+    def __init__(self, device_ids=None):
+        self._lib = load_library()
+        if device_ids is None:
+            device_ids = [0]
+        arr = (ctypes.c_int * len(device_ids))(*device_ids)
+        self._ctx = self._lib.spectre_gpu_init(len(device_ids), arr)
+        if not self._ctx:
+            raise RuntimeError(
+                "spectre_gpu_init failed (no AMD GPU?): "
+                + self._lib.spectre_gpu_last_error().decode()
+            )
+
+    def close(self):
+        if getattr(self, "_ctx", None):
+            self._lib.spectre_gpu_destroy(self._ctx)
+            self._ctx = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    def _check(self, rc, what):
+        if rc != 0:
+            raise RuntimeError(
+                f"{what} failed rc={rc}: "
+                f"{self._lib.spectre_gpu_last_error().decode()}"
+            )
+
+    # ---- host-pointer API (the halo2 seam) ----
+    def msm(self, bases: bytes | None, scalars: bytes, n: int,
+            canonical: bool = True, num_gpus: int = 1,
+            bases_id: int = 0) -> bytes:
+        out = (ctypes.c_uint8 * 64)()
+        b = (ctypes.c_uint8 * len(bases)).from_buffer_copy(bases) if bases else None
+        s = (ctypes.c_uint8 * len(scalars)).from_buffer_copy(scalars)
+        rc = self._lib.spectre_gpu_msm_g1(
+            self._ctx, bases_id, b, s, n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, num_gpus,
+            out)
+        self._check(rc, "msm_g1")
+        return bytes(out)
+
+    def ntt(self, data: bytes, log_n: int, omega: bytes, inverse: bool = False,
+            coset_gen: bytes | None = None) -> bytes:
+        buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
+        om = (ctypes.c_uint8 * 32).from_buffer_copy(omega)
+        cg = ((ctypes.c_uint8 * 32).from_buffer_copy(coset_gen)
+              if coset_gen else None)
+        rc = self._lib.spectre_gpu_ntt_fr(self._ctx, buf, log_n, om,
+                                          1 if inverse else 0, cg)
+        self._check(rc, "ntt_fr")
+        return bytes(buf)
+
+    # ---- device-resident API (device pointers, e.g. torch .data_ptr()) ----
+    def msm_device(self, d_bases: int, d_scalars: int, n: int,
+                   canonical: bool = True, dev: int = 0) -> bytes:
+        out = (ctypes.c_uint8 * 64)()
+        rc = self._lib.spectre_gpu_msm_g1_device(
+            self._ctx, dev, ctypes.c_void_p(d_bases),
+            ctypes.c_void_p(d_scalars), n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out)
+        self._check(rc, "msm_g1_device")
+        return bytes(out)
+
+    def msm_shard_device(self, d_bases: int, d_scalars: int, n: int,
+                         canonical: bool = True, dev: int = 0) -> bytes:
+        out = (ctypes.c_uint8 * PARTIALS_BYTES)()
+        rc = self._lib.spectre_gpu_msm_g1_shard_device(
+            self._ctx, dev, ctypes.c_void_p(d_bases),
+            ctypes.c_void_p(d_scalars), n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out)
+        self._check(rc, "msm_g1_shard_device")
+        return bytes(out)
+
+    def msm_shard_device_timed(self, d_bases: int, d_scalars: int, n: int,
+                               canonical: bool = True, dev: int = 0):
+        """Returns (partials_bytes, stage_ms dict) — stage timings from HIP
+        events on the library stream (see spectre_gpu.h)."""
+        out = (ctypes.c_uint8 * PARTIALS_BYTES)()
+        ms = (ctypes.c_double * 8)()
+        rc = self._lib.spectre_gpu_msm_g1_shard_device_timed(
+            self._ctx, dev, ctypes.c_void_p(d_bases),
+            ctypes.c_void_p(d_scalars), n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out, ms)
+        self._check(rc, "msm_g1_shard_device_timed")
+        stages = dict(zip(
+            ["digits", "sort", "offsets", "bucket_acc", "chunks", "reduce",
+             "total", "real_entries"], list(ms)))
+        return bytes(out), stages
+
+    def ntt_device(self, d_data: int, log_n: int, omega: bytes,
+                   inverse: bool = False, coset_gen: bytes | None = None,
+                   dev: int = 0) -> None:
+        om = (ctypes.c_uint8 * 32).from_buffer_copy(omega)
+        cg = ((ctypes.c_uint8 * 32).from_buffer_copy(coset_gen)
+              if coset_gen else None)
+        rc = self._lib.spectre_gpu_ntt_fr_device(
+            self._ctx, dev, ctypes.c_void_p(d_data), log_n, om,
+            1 if inverse else 0, cg)
+        self._check(rc, "ntt_fr_device")
+
+    # ---- device memory helpers ----
+    def malloc(self, nbytes: int, dev: int = 0) -> int:
+        p = ctypes.c_void_p()
+        self._check(self._lib.spectre_gpu_malloc(self._ctx, dev, nbytes,
+                                                 ctypes.byref(p)), "malloc")
+        return p.value
+
+    def free(self, d_ptr: int, dev: int = 0) -> None:
+        self._check(self._lib.spectre_gpu_free(self._ctx, dev,
+                                               ctypes.c_void_p(d_ptr)), "free")
+
+    def upload(self, d_dst: int, data: bytes, dev: int = 0) -> None:
+        buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
+        self._check(self._lib.spectre_gpu_upload(self._ctx, dev,
+                                                 ctypes.c_void_p(d_dst), buf,
+                                                 len(data)), "upload")
+
+    def download(self, d_src: int, nbytes: int, dev: int = 0) -> bytes:
+        buf = (ctypes.c_uint8 * nbytes)()
+        self._check(self._lib.spectre_gpu_download(self._ctx, dev, buf,
+                                                   ctypes.c_void_p(d_src),
+                                                   nbytes), "download")
+        return bytes(buf)
+
+    def synchronize(self, dev: int = 0) -> None:
+        self._check(self._lib.spectre_gpu_synchronize(self._ctx, dev), "sync")

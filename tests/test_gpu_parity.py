@@ -1,0 +1,236 @@
+"""GPU parity tests (@gpu): the HIP path vs the CPU oracle and the committed
+golden vectors, bit-exact, including the BASELINE-size property checks."""
+import hashlib
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+def hx(s):
+    return bytes.fromhex(s)
+
+
+def _omega(log_n, oracle, golden):
+    for c in golden("ntt.json")["seeded_cases"]:
+        if c["log_n"] == 12:
+            w = hx(c["omega_mont"])
+    if log_n > 12:  # lift via the 2^28 root: omega_28^(2^(28-log_n))
+        w28 = hx(ROOT28)
+        w = w28
+        for _ in range(28 - log_n):
+            w = oracle.fr_mul(w, w)
+    else:
+        for _ in range(12 - log_n):
+            w = oracle.fr_mul(w, w)
+    return w
+
+
+# 2^28 primitive root of BN254 Fr, Montgomery form (= 7^((r-1)/2^28));
+# verified against the committed 2^12 fixture root in test_root_consistency.
+ROOT28 = None
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _root28(oracle):
+    global ROOT28
+    # GENERATOR=7 canonical -> mont; exponent (r-1)/2^28
+    r = 21888242871839275222246405745257275088548364400416034343698204186575808495617
+    g7 = oracle.fr_from_canonical((7).to_bytes(32, "little"))
+    e = ((r - 1) >> 28).to_bytes(32, "little")
+    ROOT28 = oracle.fr_pow(g7, e)
+    yield
+
+
+def test_root_consistency(oracle, golden):
+    w = ROOT28
+    for _ in range(28 - 12):
+        w = oracle.fr_mul(w, w)
+    for c in golden("ntt.json")["seeded_cases"]:
+        if c["log_n"] == 12:
+            assert w.hex() == c["omega_mont"]
+
+
+# ---------------------------------------------------------------- MSM
+def test_msm_inline_golden(gpu, golden):
+    for i, c in enumerate(golden("msm.json")["inline_cases"]):
+        n = c["n"]
+        if n == 0:
+            continue  # host-pointer path handles n=0 below
+        sc = b"".join(hx(s) for s in c["scalars_canon"])
+        scm = b"".join(hx(s) for s in c["scalars_mont"])
+        bs = b"".join(hx(s) for s in c["bases"])
+        assert gpu.msm(bs, sc, n, canonical=True).hex() == c["result"], i
+        assert gpu.msm(bs, scm, n, canonical=False).hex() == c["result"], i
+
+
+def test_msm_n0(gpu):
+    assert gpu.msm(b"", b"", 0) == bytes(64)
+
+
+def test_msm_seeded_golden(gpu, oracle, golden):
+    for c in golden("msm.json")["seeded_cases"]:
+        sc, bs = oracle.gen_msm_inputs(c["n"], c["seed"])
+        assert gpu.msm(bs, sc, c["n"]).hex() == c["result"], c["n"]
+
+
+@pytest.mark.parametrize("log_n", [12, 16])
+def test_msm_vs_oracle(gpu, oracle, log_n):
+    n = 1 << log_n
+    sc, bs = oracle.gen_msm_inputs(n, 9000 + log_n, fast=True)
+    want = oracle.msm(bs, sc, n)
+    assert gpu.msm(bs, sc, n) == want
+
+
+def test_msm_2pow20_vs_oracle(gpu, oracle):
+    """Full BASELINE config[1] size, exact parity."""
+    n = 1 << 20
+    sc, bs = oracle.gen_msm_inputs(n, 42, fast=True)
+    want = oracle.msm(bs, sc, n)
+    assert gpu.msm(bs, sc, n) == want
+
+
+def test_msm_bases_cache(gpu, oracle):
+    n = 4096
+    sc, bs = oracle.gen_msm_inputs(n, 31, fast=True)
+    r1 = gpu.msm(bs, sc, n, bases_id=77)
+    r2 = gpu.msm(None, sc, n, bases_id=77)  # cached upload
+    assert r1 == r2 == oracle.msm(bs, sc, n)
+
+
+def test_msm_shard_combine_matches_direct(gpu, oracle):
+    """Two shards on one device + host combine == unsharded result — the
+    exact exchange the multi-GPU path performs (bit-identical by affine
+    canonicality)."""
+    from spectre_amd import ffi
+    n = 1 << 13
+    sc, bs = oracle.gen_msm_inputs(n, 60, fast=True)
+    direct = gpu.msm(bs, sc, n)
+    half = n // 2
+    parts = b""
+    d_b = gpu.malloc(64 * half)
+    d_s = gpu.malloc(32 * half)
+    for lo in (0, half):
+        gpu.upload(d_b, bs[64 * lo:64 * (lo + half)])
+        gpu.upload(d_s, sc[32 * lo:32 * (lo + half)])
+        parts += gpu.msm_shard_device(d_b, d_s, half)
+    gpu.free(d_b)
+    gpu.free(d_s)
+    assert ffi.combine_partials(parts, 2) == direct
+
+
+def test_msm_device_resident(gpu, oracle):
+    n = 4096
+    sc, bs = oracle.gen_msm_inputs(n, 61, fast=True)
+    d_b = gpu.malloc(64 * n)
+    d_s = gpu.malloc(32 * n)
+    gpu.upload(d_b, bs)
+    gpu.upload(d_s, sc)
+    got = gpu.msm_device(d_b, d_s, n)
+    gpu.free(d_b)
+    gpu.free(d_s)
+    assert got == oracle.msm(bs, sc, n)
+
+
+# ---------------------------------------------------------------- NTT
+def test_ntt_inline_golden(gpu, golden):
+    for c in golden("ntt.json")["inline_cases"]:
+        ln = c["log_n"]
+        inp = b"".join(hx(x) for x in c["input_mont"])
+        om = hx(c["omega_mont"])
+        assert gpu.ntt(inp, ln, om).hex() == "".join(c["ntt"]), (ln, "fwd")
+
+
+def test_ntt_inline_golden_inverse_coset(gpu, oracle, golden):
+    for c in golden("ntt.json")["inline_cases"]:
+        ln = c["log_n"]
+        inp = b"".join(hx(x) for x in c["input_mont"])
+        om = hx(c["omega_mont"])
+        out_i = gpu.ntt(inp, ln, oracle.fr_inv(om), inverse=True)
+        assert out_i.hex() == "".join(c["intt"]), (ln, "inv")
+        out_c = gpu.ntt(inp, ln, om, coset_gen=hx(c["coset_g_mont"]))
+        assert out_c.hex() == "".join(c["coset_ntt"]), (ln, "coset")
+
+
+def test_ntt_seeded_golden(gpu, oracle, golden):
+    for c in golden("ntt.json")["seeded_cases"]:
+        ln = c["log_n"]
+        inp = oracle.gen_fr_vector(1 << ln, c["seed"])
+        om = hx(c["omega_mont"])
+        assert hashlib.sha256(gpu.ntt(inp, ln, om)).hexdigest() == c["ntt_sha256"]
+        out_i = gpu.ntt(inp, ln, oracle.fr_inv(om), inverse=True)
+        assert hashlib.sha256(out_i).hexdigest() == c["intt_sha256"]
+
+
+@pytest.mark.parametrize("log_n", [13, 16, 18])
+def test_ntt_vs_oracle(gpu, oracle, golden, log_n):
+    """Two-pass GPU path vs oracle at sizes the oracle runs in seconds."""
+    n = 1 << log_n
+    a = oracle.gen_fr_vector(n, 100 + log_n)
+    w = _omega(log_n, oracle, golden)
+    assert gpu.ntt(a, log_n, w) == oracle.ntt(a, log_n, w)
+    wi = oracle.fr_inv(w)
+    assert gpu.ntt(a, log_n, wi, inverse=True) == oracle.ntt(a, log_n, wi,
+                                                             inverse=True)
+
+
+def test_ntt_coset_vs_oracle(gpu, oracle, golden):
+    log_n = 14
+    n = 1 << log_n
+    a = oracle.gen_fr_vector(n, 777)
+    w = _omega(log_n, oracle, golden)
+    g = oracle.fr_from_canonical((5).to_bytes(32, "little"))
+    assert gpu.ntt(a, log_n, w, coset_gen=g) == oracle.ntt(a, log_n, w,
+                                                           coset_gen=g)
+    gi = oracle.fr_inv(g)
+    wi = oracle.fr_inv(w)
+    assert (gpu.ntt(a, log_n, wi, inverse=True, coset_gen=gi)
+            == oracle.ntt(a, log_n, wi, inverse=True, coset_gen=gi))
+
+
+@pytest.mark.parametrize("log_n", [20, 23])
+def test_ntt_roundtrip_full_size(gpu, oracle, golden, log_n):
+    """BASELINE-size property check (oracle-free at this size):
+    iNTT(NTT(a)) == a and coset round trip, device-resident."""
+    n = 1 << log_n
+    a = oracle.gen_fr_vector(n, 4000 + log_n)
+    w = _omega(log_n, oracle, golden)
+    wi = oracle.fr_inv(w)
+    d = gpu.malloc(32 * n)
+    gpu.upload(d, a)
+    gpu.ntt_device(d, log_n, w)
+    gpu.ntt_device(d, log_n, wi, inverse=True)
+    assert gpu.download(d, 32 * n) == a
+    g = oracle.fr_from_canonical((5).to_bytes(32, "little"))
+    gpu.ntt_device(d, log_n, w, coset_gen=g)
+    gpu.ntt_device(d, log_n, wi, inverse=True, coset_gen=oracle.fr_inv(g))
+    assert gpu.download(d, 32 * n) == a
+    gpu.free(d)
+
+
+def test_ntt_full_size_spotcheck_vs_dft(gpu, oracle, golden):
+    """At 2^20, check a handful of output positions against a direct
+    DFT evaluation out[j] = sum_i a_i w^(ij) computed with Python bigints."""
+    log_n = 20
+    n = 1 << log_n
+    r = 21888242871839275222246405745257275088548364400416034343698204186575808495617
+    R = 1 << 256
+
+    def from_mont(b):
+        return int.from_bytes(b, "little") * pow(R, -1, r) % r
+
+    a = oracle.gen_fr_vector(n, 5555)
+    w = _omega(log_n, oracle, golden)
+    out = gpu.ntt(a, log_n, w)
+    wv = from_mont(w)
+    av = [from_mont(a[32 * i:32 * i + 32]) for i in range(n)]
+    import random
+    rng = random.Random(1)
+    for j in [0, 1, n - 1] + [rng.randrange(n) for _ in range(2)]:
+        wj = pow(wv, j, r)
+        acc, cur = 0, 1
+        for i in range(n):
+            acc = (acc + av[i] * cur) % r
+            cur = cur * wj % r
+        got = from_mont(out[32 * j:32 * j + 32])
+        assert got == acc, f"output {j}"
