@@ -16,8 +16,7 @@ def _omega(log_n, oracle, golden):
         if c["log_n"] == 12:
             w = hx(c["omega_mont"])
     if log_n > 12:  # lift via the 2^28 root: omega_28^(2^(28-log_n))
-        w28 = hx(ROOT28)
-        w = w28
+        w = ROOT28
         for _ in range(28 - log_n):
             w = oracle.fr_mul(w, w)
     else:
