@@ -1,0 +1,202 @@
+#!/usr/bin/env python3
+"""bench.py — BASELINE.json metric on the MI355X-native MSM/NTT backend.
+
+Workload (BASELINE.json configs[1], the single-GPU configuration the metric
+is quoted on): one "step" = one BN254 G1 Pippenger MSM over n = 2^20 seeded
+random scalars/points (synthetic, seed 42), inputs resident in HBM when the
+timed region starts. Under torchrun (one rank per GPU, RCCL) the single MSM
+is sharded contiguously across ranks; each step exchanges the 16 Jacobian
+window partial sums (1.5 KiB/rank) via all_gather over xGMI and every rank
+performs the deterministic rank-ordered combine — total work is fixed, so
+`scaling` is "strong" (the 1->8 GPU MSM scaling curve of the north star).
+
+Contract: rank 0 prints exactly ONE JSON line. The `roofline` object is
+measured live with HIP events on the library's own stream (the stream the
+kernels launch on); `traffic` is injected from the committed rocprofv3 PMC
+summary (profiles/roofline_traffic.json) when present, else null. The
+`cpu_baseline` leg times the OpenMP CPU oracle (a "port"-kind restatement of
+halo2curves best_multiexp — the reference itself is unbuildable here, see
+DESIGN.md) on the same inputs, rank 0 / N=1 only, bounded to a few reps.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, HERE)
+sys.path.insert(0, os.path.join(HERE, "oracle"))
+
+N_LOG = 20
+N = 1 << N_LOG
+SEED = 42
+HBM_PEAK_GBS = 8000.0  # MI355X spec peak, GB/s (≈6.3 TB/s achievable)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    import pywrap as oracle  # CPU oracle: cpu_baseline leg ONLY
+    from spectre_amd import SpectreGpu, ffi
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+    n_gpus = world if world > 1 else args.gpus
+    if world == 1 and args.gpus > 1:
+        log("WARNING: --gpus>1 without torchrun; using in-library sharding")
+
+    # ---- synthetic inputs (seeded, deterministic across ranks) ----
+    log(f"generating inputs n=2^{N_LOG} seed={SEED} ...")
+    t0 = time.time()
+    scalars, bases = oracle.gen_msm_inputs(N, SEED, fast=True)
+    log(f"input gen {time.time() - t0:.1f}s")
+
+    lo = N * rank // world
+    hi = N * (rank + 1) // world
+    m = hi - lo
+
+    gpu = SpectreGpu([local_rank] if world > 1 else list(range(args.gpus)))
+    # upload this rank's shard; resident in HBM before the timed region
+    d_b = gpu.malloc(64 * m)
+    d_s = gpu.malloc(32 * m)
+    gpu.upload(d_b, bases[64 * lo:64 * hi])
+    gpu.upload(d_s, scalars[32 * lo:32 * hi])
+    dev = torch.device(f"cuda:{local_rank}")
+
+    def step():
+        partials = gpu.msm_shard_device(d_b, d_s, m)
+        if dist is not None:
+            t = torch.frombuffer(bytearray(partials), dtype=torch.uint8).to(dev)
+            gath = [torch.empty_like(t) for _ in range(world)]
+            dist.all_gather(gath, t)
+            blob = b"".join(g.cpu().numpy().tobytes() for g in gath)
+            return ffi.combine_partials(blob, world)
+        return ffi.combine_partials(partials, 1)
+
+    # in-library multi-dev fallback (single process, --gpus>1, no torchrun)
+    if world == 1 and args.gpus > 1:
+        def step():  # noqa: F811
+            return gpu.msm(bases, scalars, N, num_gpus=args.gpus)
+
+    # ---- warmup ----
+    result = None
+    for _ in range(args.warmup):
+        result = step()
+    # sanity vs oracle once (outside timed region): same value every step
+    check = step()
+    assert result is None or check == result
+
+    # ---- timed region ----
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize(dev) if torch.cuda.is_available() else None
+    t0 = time.time()
+    for _ in range(args.steps):
+        result = step()
+    torch.cuda.synchronize(dev) if torch.cuda.is_available() else None
+    elapsed = time.time() - t0
+    if dist is not None:
+        te = torch.tensor([elapsed], dtype=torch.float64, device=dev)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+        dist.barrier()
+
+    value = args.steps / elapsed  # whole-job MSM(2^20)/s (total work fixed)
+    ms_per_step = elapsed * 1000.0 / args.steps
+
+    # ---- roofline: dominant kernel (bucket accumulation), HIP events ----
+    roofline = None
+    if rank == 0:
+        _, st = gpu.msm_shard_device_timed(d_b, d_s, m)
+        ent = st["real_entries"]
+        # algorithmic bytes of k_bucket_acc per launch (DESIGN.md "Roofline"):
+        # per sorted entry: 4 B index + 64 B affine point; per bucket:
+        # 2*4 B offsets read + 96 B Jacobian write.
+        algo_bytes = ent * (4 + 64) + 524288 * (8 + 96)
+        dur_s = st["bucket_acc"] / 1e3
+        achieved = algo_bytes / dur_s / 1e9
+        traffic = None
+        tpath = os.path.join(HERE, "profiles", "roofline_traffic.json")
+        if os.path.exists(tpath):
+            tj = json.load(open(tpath))
+            traffic = tj.get("k_bucket_acc_bytes_per_launch")
+        roofline = {
+            "bound": "hbm", "achieved": round(achieved, 1),
+            "peak": HBM_PEAK_GBS, "unit": "GB/s",
+            "frac": round(achieved / HBM_PEAK_GBS, 4), "traffic": traffic,
+            "kernel": "k_bucket_acc",
+            "stage_ms": {k: round(v, 3) for k, v in st.items()
+                         if k != "real_entries"},
+            "note": "modular-integer path is VALU-bound, not HBM-bound; "
+                    "achieved/peak fraction is reported against HBM as the "
+                    "only applicable roofline (no MFMA in modular arithmetic)",
+        }
+
+    # ---- CPU baseline: the OpenMP oracle, bounded sample ----
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+        log("cpu baseline (OpenMP oracle Pippenger) ...")
+        reps, tcpu = 0, 0.0
+        while reps < 3 and tcpu < 10.0:
+            c0 = time.time()
+            ref = oracle.msm(bases, scalars, N, scalars_canonical=True)
+            tcpu += time.time() - c0
+            reps += 1
+        assert ref == result, "CPU oracle and GPU disagree at bench size"
+        cpu_baseline = {
+            "value": round(reps / tcpu, 4), "unit": "MSM(2^20)/s",
+            "cores": oracle.num_threads(), "kind": "port",
+            "sample": f"{reps} x full n=2^20 MSM on host cores "
+                      f"({tcpu:.1f}s total)",
+        }
+
+    if rank == 0:
+        out = {
+            "metric": "BN254 G1 MSM n=2^20 throughput",
+            "value": round(value, 4),
+            "unit": "MSM(2^20)/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "u256",
+            "data": "synthetic",
+            "config": {
+                "workload": "msm_g1_2pow20",
+                "n": N,
+                "seed": SEED,
+                "parallelism": f"shard{n_gpus}" + ("+rccl" if world > 1 else ""),
+                "scalars": "canonical",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
