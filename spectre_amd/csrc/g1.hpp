@@ -4,6 +4,12 @@
 // with explicit degenerate-case handling. Jacobian identity: Z == 0.
 // Affine memory image = halo2curves G1Affine: x||y Montgomery Fq, 64 bytes,
 // identity encoded as (0, 0) (x=y=0 is not on y^2 = x^3 + 3).
+//
+// The accumulate primitives (g1j_dbl_ip / g1j_madd_ip / g1j_add_ip) mutate
+// their accumulator in place: on gfx950 a g1_jac is 24 VGPRs, and the
+// copy-in/copy-out style (o, p, q all distinct) makes hipcc spill to scratch
+// in the MSM kernels — the in-place forms keep the whole working set in
+// registers.
 #pragma once
 #include "ff.hpp"
 
@@ -32,38 +38,38 @@ FF_HD void g1a_neg(g1_affine& o, const g1_affine& p) {
     ff_neg<Fq>(o.y, p.y);
 }
 
-// dbl-2009-l, a = 0
-FF_HD void g1j_dbl(g1_jac& o, const g1_jac& p) {
-    if (g1j_is_inf(p)) { o = p; return; }
-    fp256 A, B, Csq, D, E, F, t;
+// p = 2p, dbl-2009-l (a = 0), in place
+FF_HD void g1j_dbl_ip(g1_jac& p) {
+    if (g1j_is_inf(p)) return;
+    fp256 A, B, C, D, E, F;
     ff_sqr<Fq>(A, p.X);
     ff_sqr<Fq>(B, p.Y);
-    ff_sqr<Fq>(Csq, B);
+    ff_sqr<Fq>(C, B);
     ff_add<Fq>(D, p.X, B);
     ff_sqr<Fq>(D, D);
     ff_sub<Fq>(D, D, A);
-    ff_sub<Fq>(D, D, Csq);
+    ff_sub<Fq>(D, D, C);
     ff_add<Fq>(D, D, D);
     ff_add<Fq>(E, A, A);
     ff_add<Fq>(E, E, A);
     ff_sqr<Fq>(F, E);
-    ff_mul<Fq>(t, p.Y, p.Z);
-    ff_add<Fq>(o.Z, t, t);
-    ff_sub<Fq>(o.X, F, D);
-    ff_sub<Fq>(o.X, o.X, D);
-    ff_sub<Fq>(t, D, o.X);
-    ff_mul<Fq>(t, E, t);
-    ff_add<Fq>(Csq, Csq, Csq);
-    ff_add<Fq>(Csq, Csq, Csq);
-    ff_add<Fq>(Csq, Csq, Csq);
-    ff_sub<Fq>(o.Y, t, Csq);
+    ff_mul<Fq>(p.Z, p.Y, p.Z);
+    ff_add<Fq>(p.Z, p.Z, p.Z);
+    ff_sub<Fq>(p.X, F, D);
+    ff_sub<Fq>(p.X, p.X, D);
+    ff_sub<Fq>(D, D, p.X);
+    ff_mul<Fq>(D, E, D);
+    ff_add<Fq>(C, C, C);
+    ff_add<Fq>(C, C, C);
+    ff_add<Fq>(C, C, C);
+    ff_sub<Fq>(p.Y, D, C);
 }
 
-// mixed add: o = p + q (q affine), madd-2007-bl + degenerate handling
-FF_HD void g1j_add_affine(g1_jac& o, const g1_jac& p, const g1_affine& q) {
-    if (g1a_is_inf(q)) { o = p; return; }
-    if (g1j_is_inf(p)) { g1j_from_affine(o, q); return; }
-    fp256 Z1Z1, U2, S2, H, HH, I, J, rr, V, t, Ynew;
+// p += q (q affine), madd-2007-bl, in place
+FF_HD void g1j_madd_ip(g1_jac& p, const g1_affine& q) {
+    if (g1a_is_inf(q)) return;
+    if (g1j_is_inf(p)) { g1j_from_affine(p, q); return; }
+    fp256 Z1Z1, U2, S2, H, HH, I, J, rr, V;
     ff_sqr<Fq>(Z1Z1, p.Z);
     ff_mul<Fq>(U2, q.x, Z1Z1);
     ff_mul<Fq>(S2, q.y, p.Z);
@@ -71,8 +77,8 @@ FF_HD void g1j_add_affine(g1_jac& o, const g1_jac& p, const g1_affine& q) {
     ff_sub<Fq>(H, U2, p.X);
     ff_sub<Fq>(rr, S2, p.Y);
     if (ff_is_zero(H)) {
-        if (ff_is_zero(rr)) { g1j_dbl(o, p); return; }
-        g1j_set_inf(o);
+        if (ff_is_zero(rr)) { g1j_dbl_ip(p); return; }
+        g1j_set_inf(p);
         return;
     }
     ff_add<Fq>(rr, rr, rr);
@@ -81,27 +87,30 @@ FF_HD void g1j_add_affine(g1_jac& o, const g1_jac& p, const g1_affine& q) {
     ff_add<Fq>(I, I, I);
     ff_mul<Fq>(J, H, I);
     ff_mul<Fq>(V, p.X, I);
-    ff_sqr<Fq>(o.X, rr);
-    ff_sub<Fq>(o.X, o.X, J);
-    ff_sub<Fq>(o.X, o.X, V);
-    ff_sub<Fq>(o.X, o.X, V);
-    ff_sub<Fq>(t, V, o.X);
-    ff_mul<Fq>(t, rr, t);
+    // Z3 = (Z1+H)^2 - Z1Z1 - HH   (consumes p.Z, H, Z1Z1, HH)
+    ff_add<Fq>(p.Z, p.Z, H);
+    ff_sqr<Fq>(p.Z, p.Z);
+    ff_sub<Fq>(p.Z, p.Z, Z1Z1);
+    ff_sub<Fq>(p.Z, p.Z, HH);
+    // X3 = rr^2 - J - 2V
+    ff_sqr<Fq>(H, rr);  // reuse H
+    ff_sub<Fq>(H, H, J);
+    ff_sub<Fq>(H, H, V);
+    ff_sub<Fq>(H, H, V);
+    // Y3 = rr*(V - X3) - 2*Y1*J
+    ff_sub<Fq>(V, V, H);
+    ff_mul<Fq>(V, rr, V);
     ff_mul<Fq>(J, p.Y, J);
     ff_add<Fq>(J, J, J);
-    ff_sub<Fq>(Ynew, t, J);
-    ff_add<Fq>(t, p.Z, H);
-    ff_sqr<Fq>(t, t);
-    ff_sub<Fq>(t, t, Z1Z1);
-    ff_sub<Fq>(o.Z, t, HH);
-    o.Y = Ynew;
+    ff_sub<Fq>(p.Y, V, J);
+    p.X = H;
 }
 
-// general add: o = p + q, add-2007-bl + degenerate handling
-FF_HD void g1j_add(g1_jac& o, const g1_jac& p, const g1_jac& q) {
-    if (g1j_is_inf(p)) { o = q; return; }
-    if (g1j_is_inf(q)) { o = p; return; }
-    fp256 Z1Z1, Z2Z2, U1, U2, S1, S2, H, I, J, rr, V, t, Ynew;
+// p += q (both Jacobian), add-2007-bl, in place
+FF_HD void g1j_add_ip(g1_jac& p, const g1_jac& q) {
+    if (g1j_is_inf(q)) return;
+    if (g1j_is_inf(p)) { p = q; return; }
+    fp256 Z1Z1, Z2Z2, U1, U2, S1, S2, H, I, J, rr, V;
     ff_sqr<Fq>(Z1Z1, p.Z);
     ff_sqr<Fq>(Z2Z2, q.Z);
     ff_mul<Fq>(U1, p.X, Z2Z2);
@@ -113,8 +122,8 @@ FF_HD void g1j_add(g1_jac& o, const g1_jac& p, const g1_jac& q) {
     ff_sub<Fq>(H, U2, U1);
     ff_sub<Fq>(rr, S2, S1);
     if (ff_is_zero(H)) {
-        if (ff_is_zero(rr)) { g1j_dbl(o, p); return; }
-        g1j_set_inf(o);
+        if (ff_is_zero(rr)) { g1j_dbl_ip(p); return; }
+        g1j_set_inf(p);
         return;
     }
     ff_add<Fq>(rr, rr, rr);
@@ -122,21 +131,38 @@ FF_HD void g1j_add(g1_jac& o, const g1_jac& p, const g1_jac& q) {
     ff_sqr<Fq>(I, I);
     ff_mul<Fq>(J, H, I);
     ff_mul<Fq>(V, U1, I);
-    ff_sqr<Fq>(o.X, rr);
-    ff_sub<Fq>(o.X, o.X, J);
-    ff_sub<Fq>(o.X, o.X, V);
-    ff_sub<Fq>(o.X, o.X, V);
-    ff_sub<Fq>(t, V, o.X);
-    ff_mul<Fq>(t, rr, t);
+    // Z3 = ((Z1+Z2)^2 - Z1Z1 - Z2Z2) * H
+    ff_add<Fq>(p.Z, p.Z, q.Z);
+    ff_sqr<Fq>(p.Z, p.Z);
+    ff_sub<Fq>(p.Z, p.Z, Z1Z1);
+    ff_sub<Fq>(p.Z, p.Z, Z2Z2);
+    ff_mul<Fq>(p.Z, p.Z, H);
+    // X3 = rr^2 - J - 2V
+    ff_sqr<Fq>(H, rr);
+    ff_sub<Fq>(H, H, J);
+    ff_sub<Fq>(H, H, V);
+    ff_sub<Fq>(H, H, V);
+    // Y3 = rr*(V - X3) - 2*S1*J
+    ff_sub<Fq>(V, V, H);
+    ff_mul<Fq>(V, rr, V);
     ff_mul<Fq>(J, S1, J);
     ff_add<Fq>(J, J, J);
-    ff_sub<Fq>(Ynew, t, J);
-    ff_add<Fq>(t, p.Z, q.Z);
-    ff_sqr<Fq>(t, t);
-    ff_sub<Fq>(t, t, Z1Z1);
-    ff_sub<Fq>(t, t, Z2Z2);
-    ff_mul<Fq>(o.Z, t, H);
-    o.Y = Ynew;
+    ff_sub<Fq>(p.Y, V, J);
+    p.X = H;
+}
+
+// ---- copy-style wrappers (host/ffi convenience) ----
+FF_HD void g1j_dbl(g1_jac& o, const g1_jac& p) {
+    o = p;
+    g1j_dbl_ip(o);
+}
+FF_HD void g1j_add_affine(g1_jac& o, const g1_jac& p, const g1_affine& q) {
+    o = p;
+    g1j_madd_ip(o, q);
+}
+FF_HD void g1j_add(g1_jac& o, const g1_jac& p, const g1_jac& q) {
+    o = p;
+    g1j_add_ip(o, q);
 }
 
 // host-side normalization (field inversion — used once per MSM result)

@@ -33,6 +33,11 @@
 #include <hipcub/hipcub.hpp>
 
 #define THREADS 256
+// Point-arithmetic kernels hold a 24-VGPR Jacobian accumulator plus formula
+// temporaries; at the default 4-waves/SIMD register budget (128 VGPR) hipcc
+// spills to scratch. Allowing 2 waves/SIMD (256 VGPR) removes the spills —
+// latency hiding comes from the serial-add structure, not occupancy.
+#define PT_KERNEL __global__ __launch_bounds__(THREADS, 2)
 
 // ---- kernel 1: signed window decomposition --------------------------------
 __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
@@ -84,7 +89,7 @@ __global__ void k_bucket_offsets(const uint32_t* __restrict__ keys,
 }
 
 // ---- kernel 4: bucket accumulation (mixed adds) ---------------------------
-__global__ void k_bucket_acc(const uint32_t* __restrict__ off,
+PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
                              const uint32_t* __restrict__ vals,
                              const g1_affine* __restrict__ bases,
                              g1_jac* __restrict__ buckets) {
@@ -97,8 +102,7 @@ __global__ void k_bucket_acc(const uint32_t* __restrict__ off,
         uint32_t v = vals[j];
         g1_affine p = bases[v & 0x7fffffffu];
         if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
-        g1_jac t = acc;
-        g1j_add_affine(acc, t, p);
+        g1j_madd_ip(acc, p);
     }
     buckets[b] = acc;
 }
@@ -108,7 +112,7 @@ __global__ void k_bucket_acc(const uint32_t* __restrict__ off,
 // Chunk covering window-local buckets [m, m+CHUNK):
 //   sum_{j=0}^{CHUNK-1} (j+1) * B[m+j]  +  m * sum_j B[m+j]
 // first term by suffix running sums, second by double-and-add (m < 2^15).
-__global__ void k_window_chunks(const g1_jac* __restrict__ buckets,
+PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
                                 g1_jac* __restrict__ out) {
     const uint32_t nchunks_pw = MSM_BPW / MSM_CHUNK;
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
@@ -116,46 +120,40 @@ __global__ void k_window_chunks(const g1_jac* __restrict__ buckets,
     uint32_t w = t / nchunks_pw;
     uint32_t cw = t % nchunks_pw;
     const g1_jac* B = buckets + (uint64_t)w * MSM_BPW + (uint64_t)cw * MSM_CHUNK;
+    // Phase order keeps at most TWO Jacobian accumulators live (3 would push
+    // past 256 VGPRs and spill): accW is parked in `out` before the
+    // double-and-add, then re-loaded for the final add.
     g1_jac accT, accW;
     g1j_set_inf(accT);
     g1j_set_inf(accW);
     for (int j = MSM_CHUNK - 1; j >= 0; j--) {
-        g1_jac x = accT;
-        g1j_add(accT, x, B[j]);
-        x = accW;
-        g1j_add(accW, x, accT);
+        g1j_add_ip(accT, B[j]);
+        g1j_add_ip(accW, accT);
     }
-    // accW += (cw*CHUNK) * accT
-    uint32_t m = cw * MSM_CHUNK;
+    out[t] = accW;  // park; accW dead
+    const uint32_t m = cw * MSM_CHUNK;
     if (m && !g1j_is_inf(accT)) {
         g1_jac acc;
         g1j_set_inf(acc);
         for (int bit = 15; bit >= 0; bit--) {
-            g1_jac x = acc;
-            g1j_dbl(acc, x);
-            if ((m >> bit) & 1) {
-                x = acc;
-                g1j_add(acc, x, accT);
-            }
+            g1j_dbl_ip(acc);
+            if ((m >> bit) & 1) g1j_add_ip(acc, accT);
         }
-        g1_jac x = accW;
-        g1j_add(accW, x, acc);
+        g1_jac w = out[t];  // accT dead; (w, acc) live
+        g1j_add_ip(w, acc);
+        out[t] = w;
     }
-    out[t] = accW;
 }
 
 // ---- kernel 6: grouped tree sum -------------------------------------------
-__global__ void k_reduce_groups(const g1_jac* __restrict__ in,
+PT_KERNEL void k_reduce_groups(const g1_jac* __restrict__ in,
                                 g1_jac* __restrict__ out, uint32_t group,
                                 uint32_t total_out) {
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= total_out) return;
     g1_jac acc;
     g1j_set_inf(acc);
-    for (uint32_t j = 0; j < group; j++) {
-        g1_jac x = acc;
-        g1j_add(acc, x, in[(uint64_t)t * group + j]);
-    }
+    for (uint32_t j = 0; j < group; j++) g1j_add_ip(acc, in[(uint64_t)t * group + j]);
     out[t] = acc;
 }
 
