@@ -22,6 +22,7 @@
 #define MSM_SKIP_KEY MSM_NB_TOTAL          // sorts after all real keys
 #define MSM_SORT_BITS 20                   // key range < 2^20
 #define MSM_CHUNK 16                       // buckets per reduction thread
+#define MSM_ACC_E 16                       // sorted entries per acc thread
 
 struct NttPlan {
     fp256* tw1 = nullptr;  // (w^{n2})^j, j < n1/2
@@ -48,6 +49,10 @@ struct DeviceState {
     size_t sort_tmp_cap = 0;
     uint32_t* d_offsets = nullptr;  // MSM_NB_TOTAL + 1
     g1_jac* d_buckets = nullptr;    // MSM_NB_TOTAL
+    uint32_t* d_firstK = nullptr;   // boundary-run side arrays (ent_cap/ACC_E)
+    uint32_t* d_lastK = nullptr;
+    g1_jac* d_firstP = nullptr;
+    g1_jac* d_lastP = nullptr;
     g1_jac* d_red = nullptr;        // reduction ping-pong (NB_TOTAL/CHUNK * 2)
     uint8_t* d_scalars = nullptr;
     size_t scal_cap = 0;  // bytes

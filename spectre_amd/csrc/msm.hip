@@ -13,11 +13,13 @@
 //                         sentinel key that sorts last and is never touched).
 //   3. k_bucket_offsets — one thread per bucket: binary-search the sorted key
 //                         stream for the bucket's segment bounds.
-//   4. k_bucket_acc     — one thread per bucket: serial Jacobian+affine mixed
-//                         adds over its segment. Deterministic by
-//                         construction (segment order is index order after
-//                         the stable sort); the affine result is canonical,
-//                         so ANY schedule yields bit-identical output bytes.
+//   4. k_bucket_acc     — equal-work partitioning: each thread owns exactly
+//                         MSM_ACC_E sorted entries (serial Jacobian+affine
+//                         mixed adds); interior runs write their bucket
+//                         exclusively, boundary runs are merged by
+//                         k_bucket_fix. Deterministic by construction, and
+//                         the affine result is canonical, so ANY schedule
+//                         yields bit-identical output bytes.
 //   5. k_window_chunks  — one thread per MSM_CHUNK consecutive buckets:
 //                         weighted suffix sum within the chunk, then one
 //                         15-bit double-and-add to lift the chunk to its
@@ -88,21 +90,86 @@ __global__ void k_bucket_offsets(const uint32_t* __restrict__ keys,
     off[b] = (uint32_t)lo;
 }
 
-// ---- kernel 4: bucket accumulation (mixed adds) ---------------------------
+// ---- kernel 4: bucket accumulation, equal-work partitioning ---------------
+// Bucket sizes are Poisson(mean ent/NB); one-thread-per-bucket makes the
+// busiest lane in a wave ~1.5x the mean (SIMT runs at the max). Instead each
+// thread owns exactly MSM_ACC_E consecutive SORTED entries: runs that start
+// AND end strictly inside the range write their bucket directly (exclusive);
+// the first and last (potentially thread-spanning) runs go to side arrays
+// keyed by bucket, merged by k_bucket_fix. Work per thread is exactly E
+// mixed adds — no imbalance — and the thread count doubles, which also
+// hides more of the dependent-multiply latency.
 PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
-                             const uint32_t* __restrict__ vals,
-                             const g1_affine* __restrict__ bases,
-                             g1_jac* __restrict__ buckets) {
+                            const uint32_t* __restrict__ keys,
+                            const uint32_t* __restrict__ vals,
+                            const g1_affine* __restrict__ bases,
+                            g1_jac* __restrict__ buckets,
+                            uint32_t* __restrict__ firstK,
+                            g1_jac* __restrict__ firstP,
+                            uint32_t* __restrict__ lastK,
+                            g1_jac* __restrict__ lastP) {
+    const uint32_t ent = off[MSM_NB_TOTAL];  // real (non-skip) entries
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t lo = (uint64_t)t * MSM_ACC_E;
+    if (lo >= ent) return;
+    uint64_t hi = lo + MSM_ACC_E;
+    if (hi > ent) hi = ent;
+    uint32_t k = keys[lo];
+    g1_jac acc;
+    g1j_set_inf(acc);
+    bool first = true;
+    for (uint64_t j = lo; j < hi; j++) {
+        uint32_t kj = keys[j];
+        if (kj != k) {  // flush completed run
+            if (first) {
+                firstK[t] = k;
+                firstP[t] = acc;
+                first = false;
+            } else if (k < MSM_NB_TOTAL) {
+                buckets[k] = acc;  // interior run: exclusive writer
+            }
+            g1j_set_inf(acc);
+            k = kj;
+        }
+        if (kj < MSM_NB_TOTAL) {
+            uint32_t v = vals[j];
+            g1_affine p = bases[v & 0x7fffffffu];
+            if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
+            g1j_madd_ip(acc, p);
+        }
+    }
+    if (first) {  // whole range is one run
+        firstK[t] = k;
+        firstP[t] = acc;
+        lastK[t] = 0xffffffffu;
+    } else {
+        lastK[t] = k;
+        lastP[t] = acc;
+    }
+}
+
+// merge boundary partials: bucket b's segment [s,e) spans threads
+// ts..te; interior-only buckets were already written by their thread.
+PT_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
+                            const uint32_t* __restrict__ firstK,
+                            const g1_jac* __restrict__ firstP,
+                            const uint32_t* __restrict__ lastK,
+                            const g1_jac* __restrict__ lastP,
+                            g1_jac* __restrict__ buckets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= MSM_NB_TOTAL) return;
     uint32_t s = off[b], e = off[b + 1];
+    if (s == e) {
+        g1j_set_inf(buckets[b]);
+        return;
+    }
+    uint32_t ts = s / MSM_ACC_E, te = (e - 1) / MSM_ACC_E;
+    if (ts == te && (s % MSM_ACC_E) && (e % MSM_ACC_E)) return;  // interior
     g1_jac acc;
     g1j_set_inf(acc);
-    for (uint32_t j = s; j < e; j++) {
-        uint32_t v = vals[j];
-        g1_affine p = bases[v & 0x7fffffffu];
-        if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
-        g1j_madd_ip(acc, p);
+    for (uint32_t t = ts; t <= te; t++) {  // ascending = sorted entry order
+        if (firstK[t] == b) g1j_add_ip(acc, firstP[t]);
+        if (lastK[t] == b) g1j_add_ip(acc, lastP[t]);
     }
     buckets[b] = acc;
 }
@@ -168,6 +235,14 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n) {
         HIP_TRY(hipMalloc(&ds.d_keys_out, ent * 4));
         HIP_TRY(hipMalloc(&ds.d_vals_in, ent * 4));
         HIP_TRY(hipMalloc(&ds.d_vals_out, ent * 4));
+        for (void* p : {(void*)ds.d_firstK, (void*)ds.d_lastK,
+                        (void*)ds.d_firstP, (void*)ds.d_lastP})
+            if (p) (void)hipFree(p);
+        const uint64_t nt = (ent + MSM_ACC_E - 1) / MSM_ACC_E;
+        HIP_TRY(hipMalloc(&ds.d_firstK, nt * 4));
+        HIP_TRY(hipMalloc(&ds.d_lastK, nt * 4));
+        HIP_TRY(hipMalloc(&ds.d_firstP, nt * sizeof(g1_jac)));
+        HIP_TRY(hipMalloc(&ds.d_lastP, nt * sizeof(g1_jac)));
         ds.ent_cap = ent;
     }
     if (!ds.d_offsets) HIP_TRY(hipMalloc(&ds.d_offsets, (MSM_NB_TOTAL + 1) * 4));
@@ -227,9 +302,14 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
                        dim3((MSM_NB_TOTAL + 1 + THREADS - 1) / THREADS),
                        dim3(THREADS), 0, st, ds.d_keys_out, ent, ds.d_offsets);
     STAMP(3);
-    hipLaunchKernelGGL(k_bucket_acc, dim3(MSM_NB_TOTAL / THREADS),
-                       dim3(THREADS), 0, st, ds.d_offsets, ds.d_vals_out,
-                       d_bases, ds.d_buckets);
+    const uint32_t nt_acc = (uint32_t)((ent + MSM_ACC_E - 1) / MSM_ACC_E);
+    hipLaunchKernelGGL(k_bucket_acc, dim3((nt_acc + THREADS - 1) / THREADS),
+                       dim3(THREADS), 0, st, ds.d_offsets, ds.d_keys_out,
+                       ds.d_vals_out, d_bases, ds.d_buckets, ds.d_firstK,
+                       ds.d_firstP, ds.d_lastK, ds.d_lastP);
+    hipLaunchKernelGGL(k_bucket_fix, dim3(MSM_NB_TOTAL / THREADS),
+                       dim3(THREADS), 0, st, ds.d_offsets, ds.d_firstK,
+                       ds.d_firstP, ds.d_lastK, ds.d_lastP, ds.d_buckets);
     STAMP(4);
     const uint32_t nchunks = MSM_NB_TOTAL / MSM_CHUNK;  // 32768
     g1_jac* red0 = ds.d_red;
