@@ -164,7 +164,13 @@ PT_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
         return;
     }
     uint32_t ts = s / MSM_ACC_E, te = (e - 1) / MSM_ACC_E;
-    if (ts == te && (s % MSM_ACC_E) && (e % MSM_ACC_E)) return;  // interior
+    // interior (already written by its exclusive thread) iff the run starts
+    // after the thread's range start AND ends before the thread's range end —
+    // note a run ending at the END OF DATA (e == ent) ended the thread's
+    // loop, so it lives in lastP, not in buckets[].
+    const uint32_t ent = off[MSM_NB_TOTAL];
+    if (ts == te && (s % MSM_ACC_E) && (e % MSM_ACC_E) && e != ent)
+        return;
     g1_jac acc;
     g1j_set_inf(acc);
     for (uint32_t t = ts; t <= te; t++) {  // ascending = sorted entry order
