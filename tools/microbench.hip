@@ -1,0 +1,84 @@
+// tools/microbench.hip — standalone probe: how many Jacobian+affine mixed
+// adds per second can one gfx950 sustain, as a function of independent
+// add-chains per thread (ILP)? Informs the bucket-accumulation design.
+// Build+run (GPU box):
+//   hipcc -O3 --offload-arch=gfx950 -I include -I spectre_amd/csrc \
+//       tools/microbench.hip -o gpurun_out/microbench && gpurun_out/microbench
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include "../spectre_amd/csrc/g1.hpp"
+
+#define ITERS 256
+
+template <int CHAINS>
+__global__ __launch_bounds__(256, 2) void k_madd_chain(const g1_affine* pts,
+                                                       g1_jac* out, int nwork) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= nwork) return;
+    g1_affine q[CHAINS];
+    g1_jac acc[CHAINS];
+    for (int c = 0; c < CHAINS; c++) {
+        q[c] = pts[(t + c * 37) % 1024];
+        g1j_from_affine(acc[c], q[c]);
+    }
+    for (int i = 0; i < ITERS; i++) {
+#pragma unroll
+        for (int c = 0; c < CHAINS; c++) g1j_madd_ip(acc[c], q[c]);
+    }
+    for (int c = 0; c < CHAINS; c++) out[t * CHAINS + c] = acc[c];
+}
+
+// pure CIOS multiply chain (field-mul latency floor)
+__global__ __launch_bounds__(256, 2) void k_mul_chain(const g1_affine* pts,
+                                                      g1_jac* out, int nwork) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= nwork) return;
+    fp256 a = pts[t % 1024].x, b = pts[t % 1024].y;
+    for (int i = 0; i < ITERS * 16; i++) ff_mul<Fq>(a, a, b);
+    out[t].X = a;
+}
+
+static double time_kernel(void (*fn)(const g1_affine*, g1_jac*, int),
+                          const g1_affine* pts, g1_jac* out, int nwork,
+                          int blocks) {
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0);
+    hipEventCreate(&e1);
+    hipLaunchKernelGGL(fn, dim3(blocks), dim3(256), 0, 0, pts, out, nwork);
+    hipDeviceSynchronize();
+    hipEventRecord(e0);
+    hipLaunchKernelGGL(fn, dim3(blocks), dim3(256), 0, 0, pts, out, nwork);
+    hipEventRecord(e1);
+    hipEventSynchronize(e1);
+    float ms;
+    hipEventElapsedTime(&ms, e0, e1);
+    hipEventDestroy(e0);
+    hipEventDestroy(e1);
+    return ms;
+}
+
+int main() {
+    // fabricate valid-shaped data (values need not be on-curve for a
+    // throughput probe; formulas don't branch on curve membership)
+    g1_affine* pts;
+    g1_jac* out;
+    hipMalloc(&pts, 1024 * sizeof(g1_affine));
+    hipMalloc(&out, (size_t)4 * 1024 * 1024 * sizeof(g1_jac));
+    hipMemset(pts, 0x5a, 1024 * sizeof(g1_affine));
+    for (int nwork : {256 * 1024, 1024 * 1024}) {
+        int blocks = nwork / 256;
+        double ms1 = time_kernel(k_madd_chain<1>, pts, out, nwork, blocks);
+        double ms2 = time_kernel(k_madd_chain<2>, pts, out, nwork / 2, blocks / 2);
+        double ms4 = time_kernel(k_madd_chain<4>, pts, out, nwork / 4, blocks / 4);
+        double madds = (double)nwork * ITERS;
+        printf("nwork=%7d  1-chain %.3f ms (%.1f M madd/s)  2-chain %.3f ms "
+               "(%.1f M madd/s)  4-chain %.3f ms (%.1f M madd/s)\n",
+               nwork, ms1, madds / ms1 / 1e3, ms2, madds / ms2 / 1e3, ms4,
+               madds / ms4 / 1e3);
+    }
+    int nwork = 1024 * 1024;
+    double msm = time_kernel(k_mul_chain, pts, out, nwork, nwork / 256);
+    printf("ff_mul chain: %.3f ms  (%.1f M mul/s)\n", msm,
+           (double)nwork * ITERS * 16 / msm / 1e3);
+    return 0;
+}
