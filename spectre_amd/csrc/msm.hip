@@ -25,7 +25,8 @@
 //                         15-bit double-and-add to lift the chunk to its
 //                         window offset. Independent across threads (no
 //                         serial suffix chain across the whole window).
-//   6. k_reduce_groups  — log-tree group sums down to MSM_NWIN window sums.
+//   6. k_window_sum     — one block per window, LDS pairwise tree to the
+//                         final MSM_NWIN window sums in a single launch.
 // Host side (ffi.cpp) finishes with 15 Horner steps (16 doubl-groups + adds)
 // and one field inversion to affine — microseconds of host work.
 //
@@ -218,16 +219,28 @@ PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
     }
 }
 
-// ---- kernel 6: grouped tree sum -------------------------------------------
-PT_KERNEL void k_reduce_groups(const g1_jac* __restrict__ in,
-                                g1_jac* __restrict__ out, uint32_t group,
-                                uint32_t total_out) {
-    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
-    if (t >= total_out) return;
+// ---- kernel 6: per-window LDS-tree sum ------------------------------------
+// One block per window sums its NB/CHUNK/NWIN chunk points: 256 threads
+// grid-stride their share, then a pairwise LDS tree (8 levels) — serial
+// depth ~16 adds in ONE launch (a 3-level launch cascade of tiny grids was
+// ~0.8 ms of pure launch+latency overhead).
+PT_KERNEL void k_window_sum(const g1_jac* __restrict__ in,
+                            g1_jac* __restrict__ out) {
+    __shared__ g1_jac lds[128];
+    const uint32_t per_win = MSM_BPW / MSM_CHUNK;
+    const uint32_t w = blockIdx.x;
+    const uint32_t t = threadIdx.x;
     g1_jac acc;
     g1j_set_inf(acc);
-    for (uint32_t j = 0; j < group; j++) g1j_add_ip(acc, in[(uint64_t)t * group + j]);
-    out[t] = acc;
+    for (uint32_t j = t; j < per_win; j += blockDim.x)
+        g1j_add_ip(acc, in[(uint64_t)w * per_win + j]);
+    for (uint32_t k = 128; k >= 1; k >>= 1) {
+        if (t >= k && t < 2 * k) lds[t - k] = acc;
+        __syncthreads();
+        if (t < k) g1j_add_ip(acc, lds[t]);
+        __syncthreads();
+    }
+    if (t == 0) out[w] = acc;
 }
 
 // ---- host orchestration ---------------------------------------------------
@@ -323,25 +336,10 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
     hipLaunchKernelGGL(k_window_chunks, dim3(nchunks / THREADS), dim3(THREADS),
                        0, st, ds.d_buckets, red0);
     STAMP(5);
-    // tree: 32768 -> 2048 -> 128 -> 16 (groups of 16, 16, 8)
-    uint32_t cur = nchunks;
-    const uint32_t groups[3] = {16, 16, 8};
-    g1_jac* in = red0;
-    g1_jac* out = red1;
-    for (int lvl = 0; lvl < 3; lvl++) {
-        uint32_t g = groups[lvl];
-        uint32_t tot = cur / g;
-        hipLaunchKernelGGL(k_reduce_groups,
-                           dim3((tot + THREADS - 1) / THREADS), dim3(THREADS),
-                           0, st, in, out, g, tot);
-        g1_jac* t = in;
-        in = out;
-        out = t;
-        cur = tot;
-    }
+    hipLaunchKernelGGL(k_window_sum, dim3(MSM_NWIN), dim3(THREADS), 0, st,
+                       red0, red1);
     STAMP(6);
-    // `in` now holds MSM_NWIN window sums
-    HIP_TRY(hipMemcpyAsync(winsums_host, in, MSM_NWIN * sizeof(g1_jac),
+    HIP_TRY(hipMemcpyAsync(winsums_host, red1, MSM_NWIN * sizeof(g1_jac),
                            hipMemcpyDeviceToHost, st));
     uint32_t ent_real = 0;
     if (stage_ms)
