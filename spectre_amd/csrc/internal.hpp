@@ -21,7 +21,7 @@
 #define MSM_NB_TOTAL (MSM_NWIN * MSM_BPW)  // 524288
 #define MSM_SKIP_KEY MSM_NB_TOTAL          // sorts after all real keys
 #define MSM_SORT_BITS 20                   // key range < 2^20
-#define MSM_CHUNK 16                       // buckets per reduction thread
+#define MSM_CHUNK 8                        // buckets per reduction thread
 #define MSM_ACC_E 32                       // sorted entries per acc thread
 
 struct NttPlan {
