@@ -24,13 +24,21 @@
 // end-to-end exactly as halo2 holds its &[Fr] slices.
 #include "internal.hpp"
 
-#define THREADS 256
+#define THREADS 256       // pow-table builder
+#define NTT_THREADS 1024  // NTT passes: 16 waves/block so one LDS-resident
+                          // block still puts 4 waves on every SIMD
 #define TW_LOW_BITS 12
 #define TW_LOW_MASK 0xfffu
 
 __device__ __forceinline__ uint32_t bitrev(uint32_t x, uint32_t bits) {
     return bits ? (__brev(x) >> (32 - bits)) : 0;
 }
+// LDS index padding: one fp256 of pad every 8 elements rotates each 8-element
+// block by 8 banks — without it, 32-B elements at power-of-two butterfly
+// strides put every lane of a ds_read_b128 lane-group on the same bank
+// (8-way serialization).
+__device__ __forceinline__ uint32_t ldsx(uint32_t i) { return i + (i >> 3); }
+#define LDS_ELEMS(L) ((L) + ((L) >> 3))
 // g^e via the 2D power table (e < n <= 2^24)
 __device__ __forceinline__ void tw_lookup(fp256& o, const fp256* T1,
                                           const fp256* T2, uint32_t e) {
@@ -57,17 +65,18 @@ __device__ void lds_dif(fp256* lds, const fp256* __restrict__ twL,
         for (uint32_t p = threadIdx.x; p < (L >> 1); p += blockDim.x) {
             uint32_t blk = p / h, j = p % h;
             uint32_t i0 = blk * 2 * h + j, i1 = i0 + h;
-            fp256 u = lds[i0], v = lds[i1], t;
-            ff_add<Fr>(lds[i0], u, v);
+            fp256 u = lds[ldsx(i0)], v = lds[ldsx(i1)], t;
+            ff_add<Fr>(lds[ldsx(i0)], u, v);
             ff_sub<Fr>(t, u, v);
-            ff_mul<Fr>(lds[i1], t, twL[(uint64_t)j * stride]);
+            ff_mul<Fr>(lds[ldsx(i1)], t, twL[(uint64_t)j * stride]);
         }
         __syncthreads();
     }
 }
 
 // pass A: column DFTs. grid.x = n2; LDS = n1 elements.
-__global__ void k_ntt_col(const fp256* __restrict__ in, fp256* __restrict__ out,
+__global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
+                          const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw1,
                           const fp256* __restrict__ T1,
                           const fp256* __restrict__ T2,
@@ -85,7 +94,7 @@ __global__ void k_ntt_col(const fp256* __restrict__ in, fp256* __restrict__ out,
             tw_lookup(f, cT1, cT2, s * n2 + c);
             ff_mul<Fr>(v, v, f);
         }
-        lds[s] = v;
+        lds[ldsx(s)] = v;
     }
     __syncthreads();
     lds_dif(lds, tw1, log_n1);
@@ -94,14 +103,15 @@ __global__ void k_ntt_col(const fp256* __restrict__ in, fp256* __restrict__ out,
         uint32_t t1 = bitrev(s, log_n1);
         fp256 f, v;
         tw_lookup(f, T1, T2, c * t1);  // c*t1 < n2*n1 = n
-        ff_mul<Fr>(v, lds[s], f);
+        ff_mul<Fr>(v, lds[ldsx(s)], f);
         out[(uint64_t)t1 * n2 + c] = v;
     }
 }
 
 // pass B: row DFTs + transposed store. grid.x = n1; LDS = n2 elements.
 // in == out is safe only when n1 == 1 (single workgroup).
-__global__ void k_ntt_row(const fp256* __restrict__ in, fp256* __restrict__ out,
+__global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
+                          const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw2,
                           const fp256* __restrict__ cT1,
                           const fp256* __restrict__ cT2, int coset_on_load,
@@ -118,13 +128,13 @@ __global__ void k_ntt_row(const fp256* __restrict__ in, fp256* __restrict__ out,
             tw_lookup(f, cT1, cT2, s);
             ff_mul<Fr>(v, v, f);
         }
-        lds[s] = v;
+        lds[ldsx(s)] = v;
     }
     __syncthreads();
     lds_dif(lds, tw2, log_n2);
     for (uint32_t s = threadIdx.x; s < n2; s += blockDim.x) {
         uint32_t t2 = bitrev(s, log_n2);
-        fp256 v = lds[s];
+        fp256 v = lds[ldsx(s)];
         if (apply_scale) ff_mul<Fr>(v, v, scale);
         if (cT1 && !coset_on_load) {  // inverse coset: g^(output index)
             fp256 f;
@@ -236,21 +246,24 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
     const fp256* inv_cT1 = (coset_gen && inverse) ? cT1 : nullptr;
     const fp256* inv_cT2 = (coset_gen && inverse) ? cT2 : nullptr;
     if (plan->k1 > 0) {
-        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(THREADS),
-                           n1 * sizeof(fp256), st, d_data, ds.d_ntt_tmp,
-                           plan->tw1, plan->twB, plan->twB + t1n, fwd_cT1,
-                           fwd_cT2, plan->k1, plan->k2);
-        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(THREADS),
-                           n2 * sizeof(fp256), st, ds.d_ntt_tmp, d_data,
-                           plan->tw2, inv_cT1, inv_cT2, /*coset_on_load=*/0,
-                           scale, inverse ? 1 : 0, plan->k1, plan->k2);
+        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(NTT_THREADS),
+                           LDS_ELEMS(n1) * sizeof(fp256), st, d_data,
+                           ds.d_ntt_tmp, plan->tw1, plan->twB,
+                           plan->twB + t1n, fwd_cT1, fwd_cT2, plan->k1,
+                           plan->k2);
+        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(NTT_THREADS),
+                           LDS_ELEMS(n2) * sizeof(fp256), st, ds.d_ntt_tmp,
+                           d_data, plan->tw2, inv_cT1, inv_cT2,
+                           /*coset_on_load=*/0, scale, inverse ? 1 : 0,
+                           plan->k1, plan->k2);
     } else {
         // single pass; forward coset applies on load, inverse on store
         const fp256* cc1 = coset_gen ? cT1 : nullptr;
         const fp256* cc2 = coset_gen ? cT2 : nullptr;
-        hipLaunchKernelGGL(k_ntt_row, dim3(1), dim3(THREADS),
-                           n2 * sizeof(fp256), st, d_data, d_data, plan->tw2,
-                           cc1, cc2, /*coset_on_load=*/inverse ? 0 : 1, scale,
+        hipLaunchKernelGGL(k_ntt_row, dim3(1), dim3(NTT_THREADS),
+                           LDS_ELEMS(n2) * sizeof(fp256), st, d_data, d_data,
+                           plan->tw2, cc1, cc2,
+                           /*coset_on_load=*/inverse ? 0 : 1, scale,
                            inverse ? 1 : 0, 0, plan->k2);
     }
     HIP_TRY(hipStreamSynchronize(st));
