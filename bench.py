@@ -129,9 +129,9 @@ def main():
         _, st = gpu.msm_shard_device_timed(d_b, d_s, m)
         ent = st["real_entries"]
         # algorithmic bytes of k_bucket_acc per launch (DESIGN.md "Roofline"):
-        # per sorted entry: 4 B index + 64 B affine point; per bucket:
-        # 2*4 B offsets read + 96 B Jacobian write.
-        algo_bytes = ent * (4 + 64) + 524288 * (8 + 96)
+        # per sorted entry: 4 B key + 4 B index + 64 B affine point; per
+        # bucket: 96 B Jacobian write (+ fixup-side arrays, excluded).
+        algo_bytes = ent * (4 + 4 + 64) + ffi.NUM_BUCKETS * 96
         dur_s = st["bucket_acc"] / 1e3
         achieved = algo_bytes / dur_s / 1e9
         traffic = None

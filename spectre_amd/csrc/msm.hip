@@ -53,22 +53,27 @@ __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
     if (!canonical) ff_from_mont<Fr>(s, s);
     uint32_t carry = 0;
     for (int w = 0; w < MSM_NWIN; w++) {
-        // 16-bit window w of the canonical scalar
-        uint32_t limb = s.l[w >> 1];
-        uint32_t d = ((w & 1) ? (limb >> 16) : (limb & 0xffffu)) + carry;
+        // c-bit window w of the canonical scalar (crosses limb boundaries)
+        const uint32_t bit0 = (uint32_t)w * MSM_WBITS;
+        const uint32_t li = bit0 >> 5, sh = bit0 & 31;
+        uint64_t pair = (uint64_t)s.l[li];
+        if (li < 7) pair |= (uint64_t)s.l[li + 1] << 32;
+        uint32_t d = ((uint32_t)(pair >> sh) & ((1u << MSM_WBITS) - 1)) + carry;
+        const uint32_t half = 1u << (MSM_WBITS - 1);
+        const uint32_t full = 1u << MSM_WBITS;
         uint32_t key, val = (uint32_t)i;
         if (d == 0) {
             carry = 0;
             key = MSM_SKIP_KEY;
-        } else if (d <= 32768u) {  // positive digit, magnitude d (2^15 kept +)
+        } else if (d <= half) {  // positive digit, magnitude d (2^(c-1) kept +)
             carry = 0;
             key = w * MSM_BPW + (d - 1);
-        } else if (d == 65536u) {  // 0xffff + carry: digit 0, carry out
+        } else if (d == full) {  // max digit + carry: digit 0, carry out
             carry = 1;
             key = MSM_SKIP_KEY;
-        } else {  // negative digit, magnitude 2^16 - d
+        } else {  // negative digit, magnitude 2^c - d
             carry = 1;
-            key = w * MSM_BPW + (65536u - d - 1);
+            key = w * MSM_BPW + (full - d - 1);
             val |= 0x80000000u;
         }
         keys[(uint64_t)w * n + i] = key;
@@ -209,7 +214,7 @@ PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
     if (m && !g1j_is_inf(accT)) {
         g1_jac acc;
         g1j_set_inf(acc);
-        for (int bit = 15; bit >= 0; bit--) {
+        for (int bit = MSM_WBITS - 1; bit >= 0; bit--) {
             g1j_dbl_ip(acc);
             if ((m >> bit) & 1) g1j_add_ip(acc, accT);
         }
@@ -326,14 +331,16 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
                        dim3(THREADS), 0, st, ds.d_offsets, ds.d_keys_out,
                        ds.d_vals_out, d_bases, ds.d_buckets, ds.d_firstK,
                        ds.d_firstP, ds.d_lastK, ds.d_lastP);
-    hipLaunchKernelGGL(k_bucket_fix, dim3(MSM_NB_TOTAL / THREADS),
+    hipLaunchKernelGGL(k_bucket_fix,
+                       dim3((MSM_NB_TOTAL + THREADS - 1) / THREADS),
                        dim3(THREADS), 0, st, ds.d_offsets, ds.d_firstK,
                        ds.d_firstP, ds.d_lastK, ds.d_lastP, ds.d_buckets);
     STAMP(4);
     const uint32_t nchunks = MSM_NB_TOTAL / MSM_CHUNK;  // 32768
     g1_jac* red0 = ds.d_red;
     g1_jac* red1 = ds.d_red + nchunks;
-    hipLaunchKernelGGL(k_window_chunks, dim3(nchunks / THREADS), dim3(THREADS),
+    hipLaunchKernelGGL(k_window_chunks,
+                       dim3((nchunks + THREADS - 1) / THREADS), dim3(THREADS),
                        0, st, ds.d_buckets, red0);
     STAMP(5);
     hipLaunchKernelGGL(k_window_sum, dim3(MSM_NWIN), dim3(THREADS), 0, st,

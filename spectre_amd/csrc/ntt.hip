@@ -169,7 +169,9 @@ static int get_plan(DeviceState& ds, const fp256& omega, uint32_t log_n,
         return 0;
     }
     NttPlan p;
-    p.k2 = log_n < 12 ? log_n : 12;
+    // balanced split (both LDS tiles small -> more blocks/CU); k2 >= k1 so
+    // the contiguous row pass gets the bigger tile
+    p.k2 = log_n <= 12 ? log_n : (log_n + 1) / 2;
     p.k1 = log_n - p.k2;
     const uint32_t n1 = 1u << p.k1, n2 = 1u << p.k2;
     const uint32_t t1n = (log_n < TW_LOW_BITS) ? (1u << log_n) : (1u << TW_LOW_BITS);
@@ -246,12 +248,14 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
     const fp256* inv_cT1 = (coset_gen && inverse) ? cT1 : nullptr;
     const fp256* inv_cT2 = (coset_gen && inverse) ? cT2 : nullptr;
     if (plan->k1 > 0) {
-        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(NTT_THREADS),
+        const uint32_t tc = n1 < NTT_THREADS ? (n1 < 64 ? 64 : n1) : NTT_THREADS;
+        const uint32_t tr = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
+        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(tc),
                            LDS_ELEMS(n1) * sizeof(fp256), st, d_data,
                            ds.d_ntt_tmp, plan->tw1, plan->twB,
                            plan->twB + t1n, fwd_cT1, fwd_cT2, plan->k1,
                            plan->k2);
-        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(NTT_THREADS),
+        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(tr),
                            LDS_ELEMS(n2) * sizeof(fp256), st, ds.d_ntt_tmp,
                            d_data, plan->tw2, inv_cT1, inv_cT2,
                            /*coset_on_load=*/0, scale, inverse ? 1 : 0,
@@ -260,7 +264,8 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         // single pass; forward coset applies on load, inverse on store
         const fp256* cc1 = coset_gen ? cT1 : nullptr;
         const fp256* cc2 = coset_gen ? cT2 : nullptr;
-        hipLaunchKernelGGL(k_ntt_row, dim3(1), dim3(NTT_THREADS),
+        const uint32_t t1p = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
+        hipLaunchKernelGGL(k_ntt_row, dim3(1), dim3(t1p),
                            LDS_ELEMS(n2) * sizeof(fp256), st, d_data, d_data,
                            plan->tw2, cc1, cc2,
                            /*coset_on_load=*/inverse ? 0 : 1, scale,

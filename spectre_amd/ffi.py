@@ -12,7 +12,9 @@ _LIB = os.path.join(_HERE, "libspectre_gpu.so")
 
 SCALARS_MONTGOMERY = 0
 SCALARS_CANONICAL = 1
-NUM_WINDOWS = 16
+WINDOW_BITS = 15  # = SPECTRE_MSM_WINDOW_BITS
+NUM_WINDOWS = 17  # = SPECTRE_MSM_NUM_WINDOWS
+NUM_BUCKETS = NUM_WINDOWS * (1 << (WINDOW_BITS - 1))
 PARTIALS_BYTES = NUM_WINDOWS * 96  # per-shard Jacobian window sums
 
 _lib = None

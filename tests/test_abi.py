@@ -50,22 +50,23 @@ def test_combine_horner_matches_oracle(oracle, golden):
     sum_w 2^(16w) * P_w — checks the host Horner + affine normalization."""
     from spectre_amd import ffi
     g1 = golden("g1.json")
+    nw, wb = ffi.NUM_WINDOWS, ffi.WINDOW_BITS
     pts = [bytes.fromhex(c["mul"]) for c in g1["mul_cases"]
            if bytes.fromhex(c["mul"]) != bytes(64)][:4]
-    for w in [0, 1, 7, 15]:
+    for w in [0, 1, 7, nw - 1]:
         partials = b"".join(
-            _jac(pts[0], oracle) if i == w else bytes(96) for i in range(16))
+            _jac(pts[0], oracle) if i == w else bytes(96) for i in range(nw))
         got = ffi.combine_partials(partials, 1)
-        k = (1 << (16 * w)) % (2**256)
+        k = (1 << (wb * w)) % (2**256)
         want = oracle.g1_mul(pts[0], k.to_bytes(32, "little"))
         assert got == want, f"window {w}"
     # multi-shard, multi-window: rank order must not matter for the value
-    partials_a = b"".join(_jac(pts[i % 4], oracle) for i in range(16))
-    partials_b = b"".join(_jac(pts[(i + 1) % 4], oracle) for i in range(16))
+    partials_a = b"".join(_jac(pts[i % 4], oracle) for i in range(nw))
+    partials_b = b"".join(_jac(pts[(i + 1) % 4], oracle) for i in range(nw))
     got = ffi.combine_partials(partials_a + partials_b, 2)
     want = bytes(64)
-    for w in range(16):
-        k = (1 << (16 * w)).to_bytes(32, "little")
+    for w in range(nw):
+        k = (1 << (wb * w)).to_bytes(32, "little")
         want = oracle.g1_add(want, oracle.g1_mul(pts[w % 4], k))
         want = oracle.g1_add(want, oracle.g1_mul(pts[(w + 1) % 4], k))
     assert got == want

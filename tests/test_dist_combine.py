@@ -27,7 +27,8 @@ def _worker(rank, world, port, point_hexes, results):
         fq_one = oracle.fq_mul(x, oracle.fq_inv(x))
         pt = bytes.fromhex(point_hexes[rank])
         partial = b"".join(
-            (pt + fq_one) if w == rank else bytes(96) for w in range(16))
+            (pt + fq_one) if w == rank
+            else bytes(96) for w in range(ffi.NUM_WINDOWS))
         local = torch.frombuffer(bytearray(partial), dtype=torch.uint8)
         gathered = [torch.zeros_like(local) for _ in range(world)]
         dist.all_gather(gathered, local)
@@ -48,8 +49,10 @@ def test_allgather_combine_gloo(oracle, golden):
     port = 29571
     mp.spawn(_worker, args=(world, port, pts, results), nprocs=world,
              join=True)
-    # expected: P0 * 2^0 + P1 * 2^16
+    # expected: P0 * 2^0 + P1 * 2^WINDOW_BITS
+    from spectre_amd import ffi
     want = oracle.g1_add(
         bytes.fromhex(pts[0]),
-        oracle.g1_mul(bytes.fromhex(pts[1]), (1 << 16).to_bytes(32, "little")))
+        oracle.g1_mul(bytes.fromhex(pts[1]),
+                      (1 << ffi.WINDOW_BITS).to_bytes(32, "little")))
     assert results[0] == results[1] == want.hex()
