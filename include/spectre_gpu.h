@@ -41,8 +41,8 @@ typedef struct spectre_gpu_ctx spectre_gpu_ctx;
 #define SPECTRE_SCALARS_MONTGOMERY 0u /* raw &[Fr] memory (the real seam) */
 #define SPECTRE_SCALARS_CANONICAL 1u  /* already to_repr()'d */
 
-#define SPECTRE_MSM_WINDOW_BITS 15u /* signed window width c */
-#define SPECTRE_MSM_NUM_WINDOWS 17u /* ceil(255/c) windows */
+#define SPECTRE_MSM_WINDOW_BITS 16u /* signed window width c */
+#define SPECTRE_MSM_NUM_WINDOWS 16u /* ceil(255/c) windows */
 
 /* ctx over `device_count` HIP devices (device_ids NULL -> 0..count-1).
  * Returns NULL on failure (no GPU, bad ids). */
