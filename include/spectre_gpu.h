@@ -43,6 +43,7 @@ typedef struct spectre_gpu_ctx spectre_gpu_ctx;
 
 #define SPECTRE_MSM_WINDOW_BITS 16u /* signed window width c */
 #define SPECTRE_MSM_NUM_WINDOWS 16u /* ceil(255/c) windows */
+#define SPECTRE_MSM_MAX_BATCH 32    /* max scalar vectors per batched call */
 
 /* ctx over `device_count` HIP devices (device_ids NULL -> 0..count-1).
  * Returns NULL on failure (no GPU, bad ids). */
@@ -63,6 +64,21 @@ int spectre_gpu_msm_g1(spectre_gpu_ctx*, uint64_t bases_id,
                        const uint8_t* bases /* n*64B */,
                        const uint8_t* scalars /* n*32B */, uint64_t n,
                        uint32_t flags, int num_gpus, uint8_t out_affine[64]);
+
+/* Batched MSM: `nbatch` scalar vectors (batch-major, nbatch*n*32 B) over ONE
+ * shared base set — the shape of create_proof's back-to-back column commits
+ * (same SRS); one fused sort/accumulate pass, bases stay cache-resident
+ * across the batch. out_affine receives nbatch results (64 B each). */
+int spectre_gpu_msm_g1_batch(spectre_gpu_ctx*, uint64_t bases_id,
+                             const uint8_t* bases /* n*64B */,
+                             const uint8_t* scalars /* nbatch*n*32B */,
+                             uint32_t nbatch, uint64_t n, uint32_t flags,
+                             uint8_t* out_affine);
+int spectre_gpu_msm_g1_batch_device(spectre_gpu_ctx*, int dev,
+                                    const void* d_bases,
+                                    const void* d_scalars, uint32_t nbatch,
+                                    uint64_t n, uint32_t flags,
+                                    uint8_t* out_affine);
 
 /* ---- MSM (device-resident, single device `dev`) ------------------------ */
 int spectre_gpu_msm_g1_device(spectre_gpu_ctx*, int dev,

@@ -171,6 +171,88 @@ int spectre_gpu_msm_g1_device(spectre_gpu_ctx* ctx, int dev,
     return 0;
 }
 
+int spectre_gpu_msm_g1_batch_device(spectre_gpu_ctx* ctx, int dev,
+                                    const void* d_bases,
+                                    const void* d_scalars, uint32_t nbatch,
+                                    uint64_t n, uint32_t flags,
+                                    uint8_t* out_affine) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    std::vector<g1_jac> wins((size_t)nbatch * MSM_NWIN);
+    int rc = msm_batch_shard_device(ctx, dev, (const g1_affine*)d_bases,
+                                    (const uint8_t*)d_scalars, nbatch, n,
+                                    flags, wins.data());
+    if (rc) return rc;
+    for (uint32_t b = 0; b < nbatch; b++)
+        winsums_to_affine(&wins[(size_t)b * MSM_NWIN], out_affine + 64 * b);
+    return 0;
+}
+
+int spectre_gpu_msm_g1_batch(spectre_gpu_ctx* ctx, uint64_t bases_id,
+                             const uint8_t* bases, const uint8_t* scalars,
+                             uint32_t nbatch, uint64_t n, uint32_t flags,
+                             uint8_t* out_affine) {
+    if (!ctx) {
+        set_err("null ctx");
+        return -1;
+    }
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    if (n == 0) {
+        memset(out_affine, 0, (size_t)64 * nbatch);
+        return 0;
+    }
+    if (!scalars) {
+        set_err("scalars is NULL");
+        return -1;
+    }
+    DeviceState& ds = ctx->devs[0];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    const uint64_t sbytes = (uint64_t)nbatch * n * 32;
+    if (ds.scal_cap < sbytes) {
+        if (ds.d_scalars) (void)hipFree(ds.d_scalars);
+        HIP_TRY(hipMalloc(&ds.d_scalars, sbytes));
+        ds.scal_cap = sbytes;
+    }
+    HIP_TRY(hipMemcpyAsync(ds.d_scalars, scalars, sbytes,
+                           hipMemcpyHostToDevice, ds.stream));
+    g1_affine* d_b = nullptr;
+    if (bases_id != 0) {
+        auto key = std::make_pair(bases_id, n * 16 + 1);
+        auto it = ds.bases_cache.find(key);
+        if (it != ds.bases_cache.end()) {
+            d_b = it->second.d_ptr;
+        } else {
+            if (!bases) {
+                set_err("bases_id %llu not cached and bases is NULL",
+                        (unsigned long long)bases_id);
+                return -4;
+            }
+            CachedBases cb;
+            cb.n = n;
+            HIP_TRY(hipMalloc(&cb.d_ptr, n * sizeof(g1_affine)));
+            HIP_TRY(hipMemcpyAsync(cb.d_ptr, bases, n * 64,
+                                   hipMemcpyHostToDevice, ds.stream));
+            ds.bases_cache.emplace(key, cb);
+            d_b = cb.d_ptr;
+        }
+    } else {
+        if (!bases) {
+            set_err("bases is NULL");
+            return -1;
+        }
+        if (ds.base_cap < n) {
+            if (ds.d_bases) (void)hipFree(ds.d_bases);
+            HIP_TRY(hipMalloc(&ds.d_bases, n * sizeof(g1_affine)));
+            ds.base_cap = n;
+        }
+        HIP_TRY(hipMemcpyAsync(ds.d_bases, bases, n * 64,
+                               hipMemcpyHostToDevice, ds.stream));
+        d_b = ds.d_bases;
+    }
+    return spectre_gpu_msm_g1_batch_device(ctx, 0, d_b, ds.d_scalars, nbatch,
+                                           n, flags, out_affine);
+}
+
 int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
                        const uint8_t* bases, const uint8_t* scalars, uint64_t n,
                        uint32_t flags, int num_gpus, uint8_t out_affine[64]) {

@@ -50,7 +50,8 @@ struct DeviceState {
     size_t ent_cap = 0;  // capacity in entries (16*n)
     void* d_sort_tmp = nullptr;
     size_t sort_tmp_cap = 0;
-    uint32_t* d_offsets = nullptr;  // MSM_NB_TOTAL + 1
+    size_t nb_cap = 0;              // bucket-array capacity (batched)
+    uint32_t* d_offsets = nullptr;  // nb_cap + 1
     g1_jac* d_buckets = nullptr;    // MSM_NB_TOTAL
     uint32_t* d_firstK = nullptr;   // boundary-run side arrays (ent_cap/ACC_E)
     uint32_t* d_lastK = nullptr;
@@ -96,6 +97,12 @@ void set_err(const char* fmt, ...);
 int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
                      const uint8_t* d_scalars, uint64_t n, uint32_t flags,
                      g1_jac* winsums_host, double* stage_ms = nullptr);
+// batch variant: nbatch scalar vectors (batch-major, nbatch*n*32 B) over one
+// shared base set; winsums_host receives nbatch*MSM_NWIN Jacobian sums.
+int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
+                           const g1_affine* d_bases, const uint8_t* d_scalars,
+                           uint32_t nbatch, uint64_t n, uint32_t flags,
+                           g1_jac* winsums_host, double* stage_ms = nullptr);
 
 // ntt.hip — in-place NTT on a device buffer (synchronizes the stream).
 int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,

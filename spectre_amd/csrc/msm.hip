@@ -2,17 +2,21 @@
 //
 // Computes what halo2curves-axiom 0.5.2 `best_multiexp` computes (the value
 // Sum_i k_i * P_i; windowing is free — see oracle/bn254.c header for the
-// parity contract). MI355X-native structure:
-//   1. k_msm_digits     — one thread per scalar: optional Montgomery->canonical
-//                         conversion in registers, signed 16-bit window
-//                         recoding, emit (bucket key, point index|sign) pairs.
-//                         Writes are window-major (w*n + i) so each window's
-//                         entries are written coalesced.
-//   2. radix sort       — hipCUB DeviceRadixSort on the 20-bit keys
-//                         (16 windows x 2^15 buckets; zero digits get a
-//                         sentinel key that sorts last and is never touched).
-//   3. k_bucket_offsets — one thread per bucket: binary-search the sorted key
-//                         stream for the bucket's segment bounds.
+// parity contract). The whole pipeline is batch-generic: `nbatch` scalar
+// vectors over ONE shared base set run as a single sort/accumulate pass
+// (create_proof commits ~17 advice columns against the same SRS back to
+// back — SURVEY.md §8f-2); nbatch = 1 is the plain best_multiexp call.
+//
+// MI355X-native structure:
+//   1. k_msm_digits     — one thread per point index: optional
+//                         Montgomery->canonical conversion in registers,
+//                         signed MSM_WBITS-bit window recoding, emit
+//                         (bucket key, point index|sign) pairs for every
+//                         (batch, window). Writes are (batch,window)-major,
+//                         so each slice is written coalesced.
+//   2. radix sort       — hipCUB DeviceRadixSort over batch*window*bucket
+//                         keys; zero digits get a sentinel key sorting last.
+//   3. k_bucket_offsets — binary-search segment bounds per bucket.
 //   4. k_bucket_acc     — equal-work partitioning: each thread owns exactly
 //                         MSM_ACC_E sorted entries (serial Jacobian+affine
 //                         mixed adds); interior runs write their bucket
@@ -20,18 +24,16 @@
 //                         k_bucket_fix. Deterministic by construction, and
 //                         the affine result is canonical, so ANY schedule
 //                         yields bit-identical output bytes.
-//   5. k_window_chunks  — one thread per MSM_CHUNK consecutive buckets:
-//                         weighted suffix sum within the chunk, then one
-//                         15-bit double-and-add to lift the chunk to its
-//                         window offset. Independent across threads (no
-//                         serial suffix chain across the whole window).
-//   6. k_window_sum     — one block per window, LDS pairwise tree to the
-//                         final MSM_NWIN window sums in a single launch.
-// Host side (ffi.cpp) finishes with 15 Horner steps (16 doubl-groups + adds)
-// and one field inversion to affine — microseconds of host work.
+//   5. k_window_chunks  — per MSM_CHUNK consecutive buckets: weighted suffix
+//                         sum + double-and-add lift to the window offset.
+//   6. k_window_sum     — one block per (batch, window), pairwise LDS tree
+//                         to the final window sums in a single launch.
+// Host side (ffi.cpp) finishes each batch with the window Horner (~255
+// doublings + adds) and one field inversion to affine.
 //
-// All field math is 8x32-limb Montgomery CIOS (ff.hpp) — VALU integer work;
-// MFMA does not apply to modular arithmetic.
+// All field math is 8x32-limb Montgomery CIOS (ff.hpp) — VALU integer work
+// (measured at the chip's v_mad_u64_u32 issue ceiling); MFMA does not apply
+// to modular arithmetic.
 #include "internal.hpp"
 #include <hipcub/hipcub.hpp>
 
@@ -39,54 +41,61 @@
 // Point-arithmetic kernels hold a 24-VGPR Jacobian accumulator plus formula
 // temporaries; at the default 4-waves/SIMD register budget (128 VGPR) hipcc
 // spills to scratch. Allowing 2 waves/SIMD (256 VGPR) removes the spills —
-// latency hiding comes from the serial-add structure, not occupancy.
+// latency hiding comes from the serial-add structure, not occupancy
+// (measured: 1 add-chain/thread already saturates the VALU issue pipe).
 #define PT_KERNEL __global__ __launch_bounds__(THREADS, 2)
 
 // ---- kernel 1: signed window decomposition --------------------------------
 __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
-                             int canonical, uint32_t* __restrict__ keys,
+                             uint32_t nbatch, int canonical,
+                             uint32_t* __restrict__ keys,
                              uint32_t* __restrict__ vals) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    fp256 s;
-    ff_from_bytes(s, scalars + 32 * i);
-    if (!canonical) ff_from_mont<Fr>(s, s);
-    uint32_t carry = 0;
-    for (int w = 0; w < MSM_NWIN; w++) {
-        // c-bit window w of the canonical scalar (crosses limb boundaries)
-        const uint32_t bit0 = (uint32_t)w * MSM_WBITS;
-        const uint32_t li = bit0 >> 5, sh = bit0 & 31;
-        uint64_t pair = (uint64_t)s.l[li];
-        if (li < 7) pair |= (uint64_t)s.l[li + 1] << 32;
-        uint32_t d = ((uint32_t)(pair >> sh) & ((1u << MSM_WBITS) - 1)) + carry;
-        const uint32_t half = 1u << (MSM_WBITS - 1);
-        const uint32_t full = 1u << MSM_WBITS;
-        uint32_t key, val = (uint32_t)i;
-        if (d == 0) {
-            carry = 0;
-            key = MSM_SKIP_KEY;
-        } else if (d <= half) {  // positive digit, magnitude d (2^(c-1) kept +)
-            carry = 0;
-            key = w * MSM_BPW + (d - 1);
-        } else if (d == full) {  // max digit + carry: digit 0, carry out
-            carry = 1;
-            key = MSM_SKIP_KEY;
-        } else {  // negative digit, magnitude 2^c - d
-            carry = 1;
-            key = w * MSM_BPW + (full - d - 1);
-            val |= 0x80000000u;
+    const uint32_t skip_key = nbatch * MSM_NB_TOTAL;
+    for (uint32_t b = 0; b < nbatch; b++) {
+        fp256 s;
+        ff_from_bytes(s, scalars + 32 * (b * n + i));
+        if (!canonical) ff_from_mont<Fr>(s, s);
+        uint32_t carry = 0;
+        for (int w = 0; w < MSM_NWIN; w++) {
+            // c-bit window w of the canonical scalar (may cross limbs)
+            const uint32_t bit0 = (uint32_t)w * MSM_WBITS;
+            const uint32_t li = bit0 >> 5, sh = bit0 & 31;
+            uint64_t pair = (uint64_t)s.l[li];
+            if (li < 7) pair |= (uint64_t)s.l[li + 1] << 32;
+            uint32_t d =
+                ((uint32_t)(pair >> sh) & ((1u << MSM_WBITS) - 1)) + carry;
+            const uint32_t half = 1u << (MSM_WBITS - 1);
+            const uint32_t full = 1u << MSM_WBITS;
+            uint32_t key, val = (uint32_t)i;
+            if (d == 0) {
+                carry = 0;
+                key = skip_key;
+            } else if (d <= half) {  // positive digit, magnitude d
+                carry = 0;
+                key = b * MSM_NB_TOTAL + w * MSM_BPW + (d - 1);
+            } else if (d == full) {  // max digit + carry: digit 0, carry out
+                carry = 1;
+                key = skip_key;
+            } else {  // negative digit, magnitude 2^c - d
+                carry = 1;
+                key = b * MSM_NB_TOTAL + w * MSM_BPW + (full - d - 1);
+                val |= 0x80000000u;
+            }
+            const uint64_t slot = ((uint64_t)b * MSM_NWIN + w) * n + i;
+            keys[slot] = key;
+            vals[slot] = val;
         }
-        keys[(uint64_t)w * n + i] = key;
-        vals[(uint64_t)w * n + i] = val;
     }
 }
 
 // ---- kernel 3: per-bucket segment bounds via binary search ----------------
 __global__ void k_bucket_offsets(const uint32_t* __restrict__ keys,
-                                 uint64_t nent, uint32_t* __restrict__ off) {
+                                 uint64_t nent, uint32_t nb_total,
+                                 uint32_t* __restrict__ off) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
-    if (b > MSM_NB_TOTAL) return;
-    // lower_bound(keys, b)
+    if (b > nb_total) return;
     uint64_t lo = 0, hi = nent;
     while (lo < hi) {
         uint64_t mid = (lo + hi) >> 1;
@@ -102,19 +111,17 @@ __global__ void k_bucket_offsets(const uint32_t* __restrict__ keys,
 // thread owns exactly MSM_ACC_E consecutive SORTED entries: runs that start
 // AND end strictly inside the range write their bucket directly (exclusive);
 // the first and last (potentially thread-spanning) runs go to side arrays
-// keyed by bucket, merged by k_bucket_fix. Work per thread is exactly E
-// mixed adds — no imbalance — and the thread count doubles, which also
-// hides more of the dependent-multiply latency.
+// keyed by bucket, merged by k_bucket_fix.
 PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
                             const uint32_t* __restrict__ keys,
                             const uint32_t* __restrict__ vals,
                             const g1_affine* __restrict__ bases,
-                            g1_jac* __restrict__ buckets,
+                            uint32_t nb_total, g1_jac* __restrict__ buckets,
                             uint32_t* __restrict__ firstK,
                             g1_jac* __restrict__ firstP,
                             uint32_t* __restrict__ lastK,
                             g1_jac* __restrict__ lastP) {
-    const uint32_t ent = off[MSM_NB_TOTAL];  // real (non-skip) entries
+    const uint32_t ent = off[nb_total];  // real (non-skip) entries
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t lo = (uint64_t)t * MSM_ACC_E;
     if (lo >= ent) return;
@@ -131,13 +138,13 @@ PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
                 firstK[t] = k;
                 firstP[t] = acc;
                 first = false;
-            } else if (k < MSM_NB_TOTAL) {
+            } else if (k < nb_total) {
                 buckets[k] = acc;  // interior run: exclusive writer
             }
             g1j_set_inf(acc);
             k = kj;
         }
-        if (kj < MSM_NB_TOTAL) {
+        if (kj < nb_total) {
             uint32_t v = vals[j];
             g1_affine p = bases[v & 0x7fffffffu];
             if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
@@ -154,16 +161,16 @@ PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
     }
 }
 
-// merge boundary partials: bucket b's segment [s,e) spans threads
-// ts..te; interior-only buckets were already written by their thread.
+// merge boundary partials: bucket b's segment [s,e) spans threads ts..te;
+// interior-only buckets were already written by their exclusive thread.
 PT_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
                             const uint32_t* __restrict__ firstK,
                             const g1_jac* __restrict__ firstP,
                             const uint32_t* __restrict__ lastK,
                             const g1_jac* __restrict__ lastP,
-                            g1_jac* __restrict__ buckets) {
+                            uint32_t nb_total, g1_jac* __restrict__ buckets) {
     uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
-    if (b >= MSM_NB_TOTAL) return;
+    if (b >= nb_total) return;
     uint32_t s = off[b], e = off[b + 1];
     if (s == e) {
         g1j_set_inf(buckets[b]);
@@ -174,7 +181,7 @@ PT_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
     // after the thread's range start AND ends before the thread's range end —
     // note a run ending at the END OF DATA (e == ent) ended the thread's
     // loop, so it lives in lastP, not in buckets[].
-    const uint32_t ent = off[MSM_NB_TOTAL];
+    const uint32_t ent = off[nb_total];
     if (ts == te && (s % MSM_ACC_E) && (e % MSM_ACC_E) && e != ent)
         return;
     g1_jac acc;
@@ -187,18 +194,19 @@ PT_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
 }
 
 // ---- kernel 5: weighted chunk reduction -----------------------------------
-// For bucket-local index j in a window, the bucket's multiplier is (j+1).
-// Chunk covering window-local buckets [m, m+CHUNK):
-//   sum_{j=0}^{CHUNK-1} (j+1) * B[m+j]  +  m * sum_j B[m+j]
-// first term by suffix running sums, second by double-and-add (m < 2^15).
+// Window-local bucket j has multiplier (j+1). For the chunk covering
+// window-local buckets [m, m+CHUNK):
+//   sum_j (j+1) B[m+j]  (suffix running sums)  +  m * sum_j B[m+j]
+// (double-and-add; m < 2^(WBITS-1)). Uniform in the flattened
+// (batch*NWIN + w) window index, so batching needs no changes here.
 PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
-                                g1_jac* __restrict__ out) {
+                               uint32_t total_chunks,
+                               g1_jac* __restrict__ out) {
     const uint32_t nchunks_pw = MSM_BPW / MSM_CHUNK;
     uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
-    if (t >= MSM_NB_TOTAL / MSM_CHUNK) return;
-    uint32_t w = t / nchunks_pw;
+    if (t >= total_chunks) return;
     uint32_t cw = t % nchunks_pw;
-    const g1_jac* B = buckets + (uint64_t)w * MSM_BPW + (uint64_t)cw * MSM_CHUNK;
+    const g1_jac* B = buckets + (uint64_t)t * MSM_CHUNK;
     // Phase order keeps at most TWO Jacobian accumulators live (3 would push
     // past 256 VGPRs and spill): accW is parked in `out` before the
     // double-and-add, then re-loaded for the final add.
@@ -225,10 +233,10 @@ PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
 }
 
 // ---- kernel 6: per-window LDS-tree sum ------------------------------------
-// One block per window sums its NB/CHUNK/NWIN chunk points: 256 threads
-// grid-stride their share, then a pairwise LDS tree (8 levels) — serial
-// depth ~16 adds in ONE launch (a 3-level launch cascade of tiny grids was
-// ~0.8 ms of pure launch+latency overhead).
+// One block per flattened (batch, window) sums its BPW/CHUNK chunk points:
+// threads grid-stride their share, then a pairwise LDS tree — serial depth
+// ~16 adds in ONE launch (a 3-level cascade of tiny grids was ~0.8 ms of
+// launch+latency overhead).
 PT_KERNEL void k_window_sum(const g1_jac* __restrict__ in,
                             g1_jac* __restrict__ out) {
     __shared__ g1_jac lds[128];
@@ -249,19 +257,19 @@ PT_KERNEL void k_window_sum(const g1_jac* __restrict__ in,
 }
 
 // ---- host orchestration ---------------------------------------------------
-static int ensure_msm_scratch(DeviceState& ds, uint64_t n) {
-    const uint64_t ent = (uint64_t)MSM_NWIN * n;
+static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
+    const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
+    const uint64_t nbt = (uint64_t)nbatch * MSM_NB_TOTAL;
     if (ds.ent_cap < ent) {
         for (void* p : {(void*)ds.d_keys_in, (void*)ds.d_keys_out,
-                        (void*)ds.d_vals_in, (void*)ds.d_vals_out})
+                        (void*)ds.d_vals_in, (void*)ds.d_vals_out,
+                        (void*)ds.d_firstK, (void*)ds.d_lastK,
+                        (void*)ds.d_firstP, (void*)ds.d_lastP})
             if (p) (void)hipFree(p);
         HIP_TRY(hipMalloc(&ds.d_keys_in, ent * 4));
         HIP_TRY(hipMalloc(&ds.d_keys_out, ent * 4));
         HIP_TRY(hipMalloc(&ds.d_vals_in, ent * 4));
         HIP_TRY(hipMalloc(&ds.d_vals_out, ent * 4));
-        for (void* p : {(void*)ds.d_firstK, (void*)ds.d_lastK,
-                        (void*)ds.d_firstP, (void*)ds.d_lastP})
-            if (p) (void)hipFree(p);
         const uint64_t nt = (ent + MSM_ACC_E - 1) / MSM_ACC_E;
         HIP_TRY(hipMalloc(&ds.d_firstK, nt * 4));
         HIP_TRY(hipMalloc(&ds.d_lastK, nt * 4));
@@ -269,17 +277,21 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n) {
         HIP_TRY(hipMalloc(&ds.d_lastP, nt * sizeof(g1_jac)));
         ds.ent_cap = ent;
     }
-    if (!ds.d_offsets) HIP_TRY(hipMalloc(&ds.d_offsets, (MSM_NB_TOTAL + 1) * 4));
-    if (!ds.d_buckets)
-        HIP_TRY(hipMalloc(&ds.d_buckets, (uint64_t)MSM_NB_TOTAL * sizeof(g1_jac)));
-    if (!ds.d_red)
+    if (ds.nb_cap < nbt) {
+        for (void* p : {(void*)ds.d_offsets, (void*)ds.d_buckets,
+                        (void*)ds.d_red})
+            if (p) (void)hipFree(p);
+        HIP_TRY(hipMalloc(&ds.d_offsets, (nbt + 1) * 4));
+        HIP_TRY(hipMalloc(&ds.d_buckets, nbt * sizeof(g1_jac)));
         HIP_TRY(hipMalloc(&ds.d_red,
-                          (uint64_t)(MSM_NB_TOTAL / MSM_CHUNK) * 2 * sizeof(g1_jac)));
+                          (nbt / MSM_CHUNK + nbatch * MSM_NWIN) * sizeof(g1_jac)));
+        ds.nb_cap = nbt;
+    }
     size_t sort_need = 0;
     (void)hipcub::DeviceRadixSort::SortPairs(nullptr, sort_need, ds.d_keys_in,
-                                       ds.d_keys_out, ds.d_vals_in,
-                                       ds.d_vals_out, (int64_t)ent, 0,
-                                       MSM_SORT_BITS, ds.stream);
+                                             ds.d_keys_out, ds.d_vals_in,
+                                             ds.d_vals_out, (int64_t)ent, 0,
+                                             32, ds.stream);
     if (ds.sort_tmp_cap < sort_need) {
         if (ds.d_sort_tmp) (void)hipFree(ds.d_sort_tmp);
         HIP_TRY(hipMalloc(&ds.d_sort_tmp, sort_need));
@@ -288,18 +300,32 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n) {
     return 0;
 }
 
-int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
-                     const uint8_t* d_scalars, uint64_t n, uint32_t flags,
-                     g1_jac* winsums_host, double* stage_ms) {
+static int end_bit_for(uint64_t max_key) {
+    int b = 0;
+    while ((1ull << b) <= max_key) b++;
+    return b;
+}
+
+int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
+                           const g1_affine* d_bases, const uint8_t* d_scalars,
+                           uint32_t nbatch, uint64_t n, uint32_t flags,
+                           g1_jac* winsums_host, double* stage_ms) {
     DeviceState& ds = ctx->devs[dev];
     HIP_TRY(hipSetDevice(ds.device_id));
+    if (nbatch == 0 || nbatch > SPECTRE_MSM_MAX_BATCH) {
+        set_err("msm: nbatch %u out of range [1,%d]", nbatch,
+                SPECTRE_MSM_MAX_BATCH);
+        return -3;
+    }
     if (n == 0) {
-        for (int w = 0; w < MSM_NWIN; w++) g1j_set_inf(winsums_host[w]);
+        for (uint32_t w = 0; w < nbatch * MSM_NWIN; w++)
+            g1j_set_inf(winsums_host[w]);
         return 0;
     }
-    int rc = ensure_msm_scratch(ds, n);
+    int rc = ensure_msm_scratch(ds, n, nbatch);
     if (rc) return rc;
-    const uint64_t ent = (uint64_t)MSM_NWIN * n;
+    const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
+    const uint32_t nbt = nbatch * MSM_NB_TOTAL;
     const int canonical = (flags & SPECTRE_SCALARS_CANONICAL) ? 1 : 0;
     hipStream_t st = ds.stream;
 
@@ -308,49 +334,49 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
     hipEvent_t ev[7];
     if (stage_ms)
         for (auto& e : ev) HIP_TRY(hipEventCreate(&e));
-#define STAMP(i)                                  \
+#define STAMP(i) \
     if (stage_ms) HIP_TRY(hipEventRecord(ev[i], st));
 
     STAMP(0);
     uint32_t blocks = (uint32_t)((n + THREADS - 1) / THREADS);
     hipLaunchKernelGGL(k_msm_digits, dim3(blocks), dim3(THREADS), 0, st,
-                       d_scalars, n, canonical, ds.d_keys_in, ds.d_vals_in);
+                       d_scalars, n, nbatch, canonical, ds.d_keys_in,
+                       ds.d_vals_in);
     STAMP(1);
     size_t tmp = ds.sort_tmp_cap;
-    (void)hipcub::DeviceRadixSort::SortPairs(ds.d_sort_tmp, tmp, ds.d_keys_in,
-                                       ds.d_keys_out, ds.d_vals_in,
-                                       ds.d_vals_out, (int64_t)ent, 0,
-                                       MSM_SORT_BITS, st);
+    (void)hipcub::DeviceRadixSort::SortPairs(
+        ds.d_sort_tmp, tmp, ds.d_keys_in, ds.d_keys_out, ds.d_vals_in,
+        ds.d_vals_out, (int64_t)ent, 0, end_bit_for(nbt), st);
     STAMP(2);
     hipLaunchKernelGGL(k_bucket_offsets,
-                       dim3((MSM_NB_TOTAL + 1 + THREADS - 1) / THREADS),
-                       dim3(THREADS), 0, st, ds.d_keys_out, ent, ds.d_offsets);
+                       dim3((nbt + 1 + THREADS - 1) / THREADS), dim3(THREADS),
+                       0, st, ds.d_keys_out, ent, nbt, ds.d_offsets);
     STAMP(3);
     const uint32_t nt_acc = (uint32_t)((ent + MSM_ACC_E - 1) / MSM_ACC_E);
     hipLaunchKernelGGL(k_bucket_acc, dim3((nt_acc + THREADS - 1) / THREADS),
                        dim3(THREADS), 0, st, ds.d_offsets, ds.d_keys_out,
-                       ds.d_vals_out, d_bases, ds.d_buckets, ds.d_firstK,
+                       ds.d_vals_out, d_bases, nbt, ds.d_buckets, ds.d_firstK,
                        ds.d_firstP, ds.d_lastK, ds.d_lastP);
-    hipLaunchKernelGGL(k_bucket_fix,
-                       dim3((MSM_NB_TOTAL + THREADS - 1) / THREADS),
+    hipLaunchKernelGGL(k_bucket_fix, dim3((nbt + THREADS - 1) / THREADS),
                        dim3(THREADS), 0, st, ds.d_offsets, ds.d_firstK,
-                       ds.d_firstP, ds.d_lastK, ds.d_lastP, ds.d_buckets);
+                       ds.d_firstP, ds.d_lastK, ds.d_lastP, nbt, ds.d_buckets);
     STAMP(4);
-    const uint32_t nchunks = MSM_NB_TOTAL / MSM_CHUNK;  // 32768
+    const uint32_t nchunks = nbt / MSM_CHUNK;
     g1_jac* red0 = ds.d_red;
     g1_jac* red1 = ds.d_red + nchunks;
     hipLaunchKernelGGL(k_window_chunks,
                        dim3((nchunks + THREADS - 1) / THREADS), dim3(THREADS),
-                       0, st, ds.d_buckets, red0);
+                       0, st, ds.d_buckets, nchunks, red0);
     STAMP(5);
-    hipLaunchKernelGGL(k_window_sum, dim3(MSM_NWIN), dim3(THREADS), 0, st,
-                       red0, red1);
+    hipLaunchKernelGGL(k_window_sum, dim3(nbatch * MSM_NWIN), dim3(THREADS),
+                       0, st, red0, red1);
     STAMP(6);
-    HIP_TRY(hipMemcpyAsync(winsums_host, red1, MSM_NWIN * sizeof(g1_jac),
+    HIP_TRY(hipMemcpyAsync(winsums_host, red1,
+                           (size_t)nbatch * MSM_NWIN * sizeof(g1_jac),
                            hipMemcpyDeviceToHost, st));
     uint32_t ent_real = 0;
     if (stage_ms)
-        HIP_TRY(hipMemcpyAsync(&ent_real, ds.d_offsets + MSM_NB_TOTAL, 4,
+        HIP_TRY(hipMemcpyAsync(&ent_real, ds.d_offsets + nbt, 4,
                                hipMemcpyDeviceToHost, st));
     HIP_TRY(hipStreamSynchronize(st));
     HIP_TRY(hipGetLastError());
@@ -367,4 +393,11 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
     }
 #undef STAMP
     return 0;
+}
+
+int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
+                     const uint8_t* d_scalars, uint64_t n, uint32_t flags,
+                     g1_jac* winsums_host, double* stage_ms) {
+    return msm_batch_shard_device(ctx, dev, d_bases, d_scalars, 1, n, flags,
+                                  winsums_host, stage_ms);
 }

@@ -48,6 +48,16 @@ def load_library() -> ctypes.CDLL:
         c.c_void_p, c.c_uint64, c.c_void_p, c.c_void_p, c.c_uint64,
         c.c_uint32, c.c_int, c.c_void_p,
     ]
+    lib.spectre_gpu_msm_g1_batch.restype = c.c_int
+    lib.spectre_gpu_msm_g1_batch.argtypes = [
+        c.c_void_p, c.c_uint64, c.c_void_p, c.c_void_p, c.c_uint32,
+        c.c_uint64, c.c_uint32, c.c_void_p,
+    ]
+    lib.spectre_gpu_msm_g1_batch_device.restype = c.c_int
+    lib.spectre_gpu_msm_g1_batch_device.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint32, c.c_uint64,
+        c.c_uint32, c.c_void_p,
+    ]
     lib.spectre_gpu_msm_g1_device.restype = c.c_int
     lib.spectre_gpu_msm_g1_device.argtypes = [
         c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
@@ -154,6 +164,32 @@ class SpectreGpu:
             out)
         self._check(rc, "msm_g1")
         return bytes(out)
+
+    def msm_batch(self, bases: bytes | None, scalars: bytes, nbatch: int,
+                  n: int, canonical: bool = True, bases_id: int = 0) -> list[bytes]:
+        """nbatch MSMs over one shared base set (batch-major scalars)."""
+        assert len(scalars) == nbatch * n * 32
+        out = (ctypes.c_uint8 * (64 * nbatch))()
+        b = (ctypes.c_uint8 * len(bases)).from_buffer_copy(bases) if bases else None
+        s = (ctypes.c_uint8 * len(scalars)).from_buffer_copy(scalars)
+        rc = self._lib.spectre_gpu_msm_g1_batch(
+            self._ctx, bases_id, b, s, nbatch, n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out)
+        self._check(rc, "msm_g1_batch")
+        raw = bytes(out)
+        return [raw[64 * i:64 * (i + 1)] for i in range(nbatch)]
+
+    def msm_batch_device(self, d_bases: int, d_scalars: int, nbatch: int,
+                         n: int, canonical: bool = True,
+                         dev: int = 0) -> list[bytes]:
+        out = (ctypes.c_uint8 * (64 * nbatch))()
+        rc = self._lib.spectre_gpu_msm_g1_batch_device(
+            self._ctx, dev, ctypes.c_void_p(d_bases),
+            ctypes.c_void_p(d_scalars), nbatch, n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out)
+        self._check(rc, "msm_g1_batch_device")
+        raw = bytes(out)
+        return [raw[64 * i:64 * (i + 1)] for i in range(nbatch)]
 
     def ntt(self, data: bytes, log_n: int, omega: bytes, inverse: bool = False,
             coset_gen: bytes | None = None) -> bytes:

@@ -118,6 +118,26 @@ def test_msm_shard_combine_matches_direct(gpu, oracle):
     assert ffi.combine_partials(parts, 2) == direct
 
 
+def test_msm_batch_matches_individual(gpu, oracle):
+    """Batched MSM (shared bases, fused pipeline) must equal per-vector
+    results bit-for-bit."""
+    n, nbatch = 2048, 5
+    _, bs = oracle.gen_msm_inputs(n, 70, fast=True)
+    scal = b""
+    singles = []
+    for b in range(nbatch):
+        sc, _ = oracle.gen_msm_inputs(n, 71 + b, fast=True)
+        scal += sc
+        singles.append(oracle.msm(bs, sc, n))
+    got = gpu.msm_batch(bs, scal, nbatch, n)
+    assert got == singles
+    # edge: batch with an all-zero vector and a duplicate vector
+    scal2 = bytes(32 * n) + scal[:32 * n] + scal[:32 * n]
+    got2 = gpu.msm_batch(bs, scal2, 3, n)
+    assert got2[0] == bytes(64)
+    assert got2[1] == got2[2] == singles[0]
+
+
 def test_msm_device_resident(gpu, oracle):
     n = 4096
     sc, bs = oracle.gen_msm_inputs(n, 61, fast=True)
