@@ -63,7 +63,6 @@ void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
         for (void* p :
              {(void*)ds.d_keys_in, (void*)ds.d_keys_out, (void*)ds.d_vals_in,
               (void*)ds.d_vals_out, (void*)ds.d_sort_tmp, (void*)ds.d_offsets,
-              (void*)ds.d_counts,
               (void*)ds.d_buckets, (void*)ds.d_firstK, (void*)ds.d_lastK,
               (void*)ds.d_firstP, (void*)ds.d_lastP,
               (void*)ds.d_red, (void*)ds.d_scalars,

@@ -52,7 +52,6 @@ struct DeviceState {
     size_t sort_tmp_cap = 0;
     size_t nb_cap = 0;              // bucket-array capacity (batched)
     uint32_t* d_offsets = nullptr;  // nb_cap + 1
-    uint32_t* d_counts = nullptr;   // histogram / scatter cursors
     g1_jac* d_buckets = nullptr;    // MSM_NB_TOTAL
     uint32_t* d_firstK = nullptr;   // boundary-run side arrays (ent_cap/ACC_E)
     uint32_t* d_lastK = nullptr;

@@ -14,16 +14,9 @@
 //                         (bucket key, point index|sign) pairs for every
 //                         (batch, window). Writes are (batch,window)-major,
 //                         so each slice is written coalesced.
-//   2. counting sort    — the key space is only nbatch*2^19, so a fused
-//                         histogram (inside k_msm_digits, atomicAdd) +
-//                         hipCUB ExclusiveSum scan + k_scatter replaces a
-//                         general radix sort at ~1/4 the traffic, and the
-//                         scan output IS the bucket-offsets array (zero
-//                         digits are never materialized). Scatter order
-//                         within a bucket is nondeterministic (atomic
-//                         cursors) — harmless: group addition is exact, and
-//                         the canonical affine output is order-independent
-//                         (pinned by tests at every size).
+//   2. radix sort       — hipCUB DeviceRadixSort over batch*window*bucket
+//                         keys; zero digits get a sentinel key sorting last.
+//   3. k_bucket_offsets — binary-search segment bounds per bucket.
 //   4. k_bucket_acc     — equal-work partitioning: each thread owns exactly
 //                         MSM_ACC_E sorted entries (serial Jacobian+affine
 //                         mixed adds); interior runs write their bucket
@@ -44,7 +37,6 @@
 #include "internal.hpp"
 #include <hipcub/hipcub.hpp>
 
-
 #define THREADS 256
 // Point-arithmetic kernels hold a 24-VGPR Jacobian accumulator plus formula
 // temporaries; at the default 4-waves/SIMD register budget (128 VGPR) hipcc
@@ -57,8 +49,7 @@
 __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
                              uint32_t nbatch, int canonical,
                              uint32_t* __restrict__ keys,
-                             uint32_t* __restrict__ vals,
-                             uint32_t* __restrict__ counts) {
+                             uint32_t* __restrict__ vals) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     const uint32_t skip_key = nbatch * MSM_NB_TOTAL;
@@ -95,24 +86,23 @@ __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
             const uint64_t slot = ((uint64_t)b * MSM_NWIN + w) * n + i;
             keys[slot] = key;
             vals[slot] = val;
-            if (key != skip_key) atomicAdd(&counts[key], 1u);
         }
     }
 }
 
-// ---- kernel 3: scatter into bucket segments (cursors = scanned counts) ----
-__global__ void k_scatter(const uint32_t* __restrict__ keys_in,
-                          const uint32_t* __restrict__ vals_in, uint64_t nent,
-                          uint32_t skip_key, uint32_t* __restrict__ cursors,
-                          uint32_t* __restrict__ keys_out,
-                          uint32_t* __restrict__ vals_out) {
-    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= nent) return;
-    uint32_t k = keys_in[i];
-    if (k >= skip_key) return;  // zero digits never materialize
-    uint32_t pos = atomicAdd(&cursors[k], 1u);
-    keys_out[pos] = k;
-    vals_out[pos] = vals_in[i];
+// ---- kernel 3: per-bucket segment bounds via binary search ----------------
+__global__ void k_bucket_offsets(const uint32_t* __restrict__ keys,
+                                 uint64_t nent, uint32_t nb_total,
+                                 uint32_t* __restrict__ off) {
+    uint32_t b = blockIdx.x * blockDim.x + threadIdx.x;
+    if (b > nb_total) return;
+    uint64_t lo = 0, hi = nent;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1;
+        if (keys[mid] < b) lo = mid + 1;
+        else hi = mid;
+    }
+    off[b] = (uint32_t)lo;
 }
 
 // ---- kernel 4: bucket accumulation, equal-work partitioning ---------------
@@ -288,26 +278,32 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
         ds.ent_cap = ent;
     }
     if (ds.nb_cap < nbt) {
-        for (void* p : {(void*)ds.d_offsets, (void*)ds.d_counts,
-                        (void*)ds.d_buckets, (void*)ds.d_red})
+        for (void* p : {(void*)ds.d_offsets, (void*)ds.d_buckets,
+                        (void*)ds.d_red})
             if (p) (void)hipFree(p);
         HIP_TRY(hipMalloc(&ds.d_offsets, (nbt + 1) * 4));
-        HIP_TRY(hipMalloc(&ds.d_counts, (nbt + 1) * 4));
         HIP_TRY(hipMalloc(&ds.d_buckets, nbt * sizeof(g1_jac)));
         HIP_TRY(hipMalloc(&ds.d_red,
                           (nbt / MSM_CHUNK + nbatch * MSM_NWIN) * sizeof(g1_jac)));
         ds.nb_cap = nbt;
     }
-    size_t scan_need = 0;
-    (void)hipcub::DeviceScan::ExclusiveSum(nullptr, scan_need, ds.d_counts,
-                                           ds.d_offsets, (int64_t)(nbt + 1),
-                                           ds.stream);
-    if (ds.sort_tmp_cap < scan_need) {
+    size_t sort_need = 0;
+    (void)hipcub::DeviceRadixSort::SortPairs(nullptr, sort_need, ds.d_keys_in,
+                                             ds.d_keys_out, ds.d_vals_in,
+                                             ds.d_vals_out, (int64_t)ent, 0,
+                                             32, ds.stream);
+    if (ds.sort_tmp_cap < sort_need) {
         if (ds.d_sort_tmp) (void)hipFree(ds.d_sort_tmp);
-        HIP_TRY(hipMalloc(&ds.d_sort_tmp, scan_need));
-        ds.sort_tmp_cap = scan_need;
+        HIP_TRY(hipMalloc(&ds.d_sort_tmp, sort_need));
+        ds.sort_tmp_cap = sort_need;
     }
     return 0;
+}
+
+static int end_bit_for(uint64_t max_key) {
+    int b = 0;
+    while ((1ull << b) <= max_key) b++;
+    return b;
 }
 
 int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
@@ -342,24 +338,19 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
     if (stage_ms) HIP_TRY(hipEventRecord(ev[i], st));
 
     STAMP(0);
-    HIP_TRY(hipMemsetAsync(ds.d_counts, 0, (nbt + 1) * 4, st));
     uint32_t blocks = (uint32_t)((n + THREADS - 1) / THREADS);
     hipLaunchKernelGGL(k_msm_digits, dim3(blocks), dim3(THREADS), 0, st,
                        d_scalars, n, nbatch, canonical, ds.d_keys_in,
-                       ds.d_vals_in, ds.d_counts);
+                       ds.d_vals_in);
     STAMP(1);
     size_t tmp = ds.sort_tmp_cap;
-    (void)hipcub::DeviceScan::ExclusiveSum(ds.d_sort_tmp, tmp, ds.d_counts,
-                                           ds.d_offsets, (int64_t)(nbt + 1),
-                                           st);
+    (void)hipcub::DeviceRadixSort::SortPairs(
+        ds.d_sort_tmp, tmp, ds.d_keys_in, ds.d_keys_out, ds.d_vals_in,
+        ds.d_vals_out, (int64_t)ent, 0, end_bit_for(nbt), st);
     STAMP(2);
-    // offsets double as scatter cursors (consumed); re-derive via memcpy
-    HIP_TRY(hipMemcpyAsync(ds.d_counts, ds.d_offsets, (nbt + 1) * 4,
-                           hipMemcpyDeviceToDevice, st));
-    hipLaunchKernelGGL(k_scatter,
-                       dim3((uint32_t)((ent + THREADS - 1) / THREADS)),
-                       dim3(THREADS), 0, st, ds.d_keys_in, ds.d_vals_in, ent,
-                       nbt, ds.d_counts, ds.d_keys_out, ds.d_vals_out);
+    hipLaunchKernelGGL(k_bucket_offsets,
+                       dim3((nbt + 1 + THREADS - 1) / THREADS), dim3(THREADS),
+                       0, st, ds.d_keys_out, ent, nbt, ds.d_offsets);
     STAMP(3);
     const uint32_t nt_acc = (uint32_t)((ent + MSM_ACC_E - 1) / MSM_ACC_E);
     hipLaunchKernelGGL(k_bucket_acc, dim3((nt_acc + THREADS - 1) / THREADS),

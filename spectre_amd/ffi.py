@@ -235,7 +235,7 @@ class SpectreGpu:
             SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out, ms)
         self._check(rc, "msm_g1_shard_device_timed")
         stages = dict(zip(
-            ["digits_hist", "scan", "scatter", "bucket_acc", "chunks", "reduce",
+            ["digits", "sort", "offsets", "bucket_acc", "chunks", "reduce",
              "total", "real_entries"], list(ms)))
         return bytes(out), stages
 
