@@ -33,12 +33,27 @@
 __device__ __forceinline__ uint32_t bitrev(uint32_t x, uint32_t bits) {
     return bits ? (__brev(x) >> (32 - bits)) : 0;
 }
-// LDS index padding: one fp256 of pad every 8 elements rotates each 8-element
-// block by 8 banks — without it, 32-B elements at power-of-two butterfly
-// strides put every lane of a ds_read_b128 lane-group on the same bank
-// (8-way serialization).
-__device__ __forceinline__ uint32_t ldsx(uint32_t i) { return i + (i >> 3); }
-#define LDS_ELEMS(L) ((L) + ((L) >> 3))
+// LDS layout: a 32-B element spans 8 banks, so a 16-lane ds_read_b128 group
+// covers only 8 bank-octets — inherent 2-8x conflict at power-of-two
+// butterfly strides (measured SQ_LDS_BANK_CONFLICT/IDX_ACTIVE = 0.85 with a
+// single fp256 array). Instead each element is stored as TWO 16-B halves in
+// separate regions (lo at [i], hi at [H+i]): consecutive elements stride 4
+// banks and one lane group covers all 64 banks.
+#define LDS_BYTES(L) ((uint32_t)(L) * 32u)
+__device__ __forceinline__ void lds_ld(const uint4* lds4, uint32_t H,
+                                       uint32_t i, fp256& o) {
+    uint4 lo = lds4[i], hi = lds4[H + i];
+    memcpy(&o.l[0], &lo, 16);
+    memcpy(&o.l[4], &hi, 16);
+}
+__device__ __forceinline__ void lds_st(uint4* lds4, uint32_t H, uint32_t i,
+                                       const fp256& v) {
+    uint4 lo, hi;
+    memcpy(&lo, &v.l[0], 16);
+    memcpy(&hi, &v.l[4], 16);
+    lds4[i] = lo;
+    lds4[H + i] = hi;
+}
 // g^e via the 2D power table (e < n <= 2^24)
 __device__ __forceinline__ void tw_lookup(fp256& o, const fp256* T1,
                                           const fp256* T2, uint32_t e) {
@@ -57,7 +72,7 @@ __global__ void k_pow_table(fp256 base, fp256* __restrict__ out,
 
 // in-LDS DIF butterflies over L = 2^logL elements; twL[j] = (root)^j, j < L/2.
 // On exit lds[s] holds DFT output index bitrev(s, logL).
-__device__ void lds_dif(fp256* lds, const fp256* __restrict__ twL,
+__device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
                         uint32_t logL) {
     const uint32_t L = 1u << logL;
     for (uint32_t h = L >> 1; h >= 1; h >>= 1) {
@@ -65,10 +80,14 @@ __device__ void lds_dif(fp256* lds, const fp256* __restrict__ twL,
         for (uint32_t p = threadIdx.x; p < (L >> 1); p += blockDim.x) {
             uint32_t blk = p / h, j = p % h;
             uint32_t i0 = blk * 2 * h + j, i1 = i0 + h;
-            fp256 u = lds[ldsx(i0)], v = lds[ldsx(i1)], t;
-            ff_add<Fr>(lds[ldsx(i0)], u, v);
+            fp256 u, v, t, w;
+            lds_ld(lds4, H, i0, u);
+            lds_ld(lds4, H, i1, v);
+            ff_add<Fr>(w, u, v);
+            lds_st(lds4, H, i0, w);
             ff_sub<Fr>(t, u, v);
-            ff_mul<Fr>(lds[ldsx(i1)], t, twL[(uint64_t)j * stride]);
+            ff_mul<Fr>(t, t, twL[(uint64_t)j * stride]);
+            lds_st(lds4, H, i1, t);
         }
         __syncthreads();
     }
@@ -83,9 +102,10 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
                           const fp256* __restrict__ cT1,
                           const fp256* __restrict__ cT2, uint32_t log_n1,
                           uint32_t log_n2) {
-    extern __shared__ fp256 lds[];
+    extern __shared__ uint4 lds4[];
     const uint32_t n1 = 1u << log_n1;
     const uint32_t n2 = 1u << log_n2;
+    const uint32_t H = n1;
     const uint32_t c = blockIdx.x;
     for (uint32_t s = threadIdx.x; s < n1; s += blockDim.x) {
         fp256 v = in[(uint64_t)s * n2 + c];
@@ -94,16 +114,18 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
             tw_lookup(f, cT1, cT2, s * n2 + c);
             ff_mul<Fr>(v, v, f);
         }
-        lds[ldsx(s)] = v;
+        lds_st(lds4, H, s, v);
     }
     __syncthreads();
-    lds_dif(lds, tw1, log_n1);
+    lds_dif(lds4, H, tw1, log_n1);
     // store with inter-pass twiddle omega^(c * t1), t1 = bitrev(s)
     for (uint32_t s = threadIdx.x; s < n1; s += blockDim.x) {
         uint32_t t1 = bitrev(s, log_n1);
         fp256 f, v;
         tw_lookup(f, T1, T2, c * t1);  // c*t1 < n2*n1 = n
-        ff_mul<Fr>(v, lds[ldsx(s)], f);
+        fp256 x;
+        lds_ld(lds4, H, s, x);
+        ff_mul<Fr>(v, x, f);
         out[(uint64_t)t1 * n2 + c] = v;
     }
 }
@@ -117,9 +139,10 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
                           const fp256* __restrict__ cT2, int coset_on_load,
                           fp256 scale, int apply_scale, uint32_t log_n1,
                           uint32_t log_n2) {
-    extern __shared__ fp256 lds[];
+    extern __shared__ uint4 lds4[];
     const uint32_t n1 = 1u << log_n1;
     const uint32_t n2 = 1u << log_n2;
+    const uint32_t H = n2;
     const uint32_t r = blockIdx.x;
     for (uint32_t s = threadIdx.x; s < n2; s += blockDim.x) {
         fp256 v = in[(uint64_t)r * n2 + s];
@@ -128,13 +151,14 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
             tw_lookup(f, cT1, cT2, s);
             ff_mul<Fr>(v, v, f);
         }
-        lds[ldsx(s)] = v;
+        lds_st(lds4, H, s, v);
     }
     __syncthreads();
-    lds_dif(lds, tw2, log_n2);
+    lds_dif(lds4, H, tw2, log_n2);
     for (uint32_t s = threadIdx.x; s < n2; s += blockDim.x) {
         uint32_t t2 = bitrev(s, log_n2);
-        fp256 v = lds[ldsx(s)];
+        fp256 v;
+        lds_ld(lds4, H, s, v);
         if (apply_scale) ff_mul<Fr>(v, v, scale);
         if (cT1 && !coset_on_load) {  // inverse coset: g^(output index)
             fp256 f;
@@ -251,12 +275,12 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         const uint32_t tc = n1 < NTT_THREADS ? (n1 < 64 ? 64 : n1) : NTT_THREADS;
         const uint32_t tr = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
         hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(tc),
-                           LDS_ELEMS(n1) * sizeof(fp256), st, d_data,
+                           LDS_BYTES(n1), st, d_data,
                            ds.d_ntt_tmp, plan->tw1, plan->twB,
                            plan->twB + t1n, fwd_cT1, fwd_cT2, plan->k1,
                            plan->k2);
         hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(tr),
-                           LDS_ELEMS(n2) * sizeof(fp256), st, ds.d_ntt_tmp,
+                           LDS_BYTES(n2), st, ds.d_ntt_tmp,
                            d_data, plan->tw2, inv_cT1, inv_cT2,
                            /*coset_on_load=*/0, scale, inverse ? 1 : 0,
                            plan->k1, plan->k2);
@@ -266,7 +290,7 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         const fp256* cc2 = coset_gen ? cT2 : nullptr;
         const uint32_t t1p = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
         hipLaunchKernelGGL(k_ntt_row, dim3(1), dim3(t1p),
-                           LDS_ELEMS(n2) * sizeof(fp256), st, d_data, d_data,
+                           LDS_BYTES(n2), st, d_data, d_data,
                            plan->tw2, cc1, cc2,
                            /*coset_on_load=*/inverse ? 0 : 1, scale,
                            inverse ? 1 : 0, 0, plan->k2);
