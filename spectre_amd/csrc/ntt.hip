@@ -93,38 +93,9 @@ __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
     }
 }
 
-// 32x32-element tiled transpose (out[c*R+r] = in[r*C+c]): both global sides
-// fully coalesced; the LDS tile uses 16B halves with +1 padding. Three of
-// these (0.1 ms each at 2^23) replace the three strided DFT-pass sides,
-// which cost ~0.3 ms each.
-#define TT 32
-__global__ __launch_bounds__(256) void k_transpose(const fp256* __restrict__ in,
-                                                   fp256* __restrict__ out,
-                                                   uint32_t R, uint32_t C) {
-    __shared__ uint4 tlo[TT][TT + 1];
-    __shared__ uint4 thi[TT][TT + 1];
-    const uint32_t tr = blockIdx.y * TT;
-    const uint32_t tc = blockIdx.x * TT;
-    const uint32_t lx = threadIdx.x & 31;
-    const uint32_t ly = threadIdx.x >> 5;  // 0..7
-    for (uint32_t k = 0; k < TT; k += 8) {
-        const uint4* src =
-            (const uint4*)&in[(uint64_t)(tr + ly + k) * C + (tc + lx)];
-        tlo[ly + k][lx] = src[0];
-        thi[ly + k][lx] = src[1];
-    }
-    __syncthreads();
-    for (uint32_t k = 0; k < TT; k += 8) {
-        uint4* dst = (uint4*)&out[(uint64_t)(tc + ly + k) * R + (tr + lx)];
-        dst[0] = tlo[lx][ly + k];
-        dst[1] = thi[lx][ly + k];
-    }
-}
-
-// pass A: column DFTs over the TRANSPOSED intermediate (tmp is n2 x n1
-// row-major, so each block's column is contiguous). grid.x = n2; in place.
+// pass A: column DFTs. grid.x = n2; LDS = n1 elements.
 __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
-                          fp256* __restrict__ data /* n2 x n1, in place */,
+                          const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw1,
                           const fp256* __restrict__ T1,
                           const fp256* __restrict__ T2,
@@ -136,10 +107,9 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
     const uint32_t n2 = 1u << log_n2;
     const uint32_t H = n1;
     const uint32_t c = blockIdx.x;
-    fp256* row = data + (uint64_t)c * n1;
     for (uint32_t s = threadIdx.x; s < n1; s += blockDim.x) {
-        fp256 v = row[s];
-        if (cT1) {  // forward coset: multiply by g^(original global index)
+        fp256 v = in[(uint64_t)s * n2 + c];
+        if (cT1) {  // forward coset: multiply by g^(global index)
             fp256 f;
             tw_lookup(f, cT1, cT2, s * n2 + c);
             ff_mul<Fr>(v, v, f);
@@ -151,17 +121,17 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
     // store with inter-pass twiddle omega^(c * t1), t1 = bitrev(s)
     for (uint32_t s = threadIdx.x; s < n1; s += blockDim.x) {
         uint32_t t1 = bitrev(s, log_n1);
-        fp256 f, v, x;
+        fp256 f, v;
         tw_lookup(f, T1, T2, c * t1);  // c*t1 < n2*n1 = n
+        fp256 x;
         lds_ld(lds4, H, s, x);
         ff_mul<Fr>(v, x, f);
-        row[t1] = v;
+        out[(uint64_t)t1 * n2 + c] = v;
     }
 }
 
-// pass B: row DFTs, contiguous store out[r*n2 + t2] (the final transpose to
-// out[t2*n1 + r] is a separate coalesced k_transpose). grid.x = n1.
-// in == out is safe per block (barrier between load and store).
+// pass B: row DFTs + transposed store. grid.x = n1; LDS = n2 elements.
+// in == out is safe only when n1 == 1 (single workgroup).
 __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
                           const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw2,
@@ -195,7 +165,7 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
             tw_lookup(f, cT1, cT2, t2 * n1 + r);
             ff_mul<Fr>(v, v, f);
         }
-        out[(uint64_t)r * n2 + t2] = v;
+        out[(uint64_t)t2 * n1 + r] = v;
     }
 }
 
@@ -304,21 +274,16 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
     if (plan->k1 > 0) {
         const uint32_t tc = n1 < NTT_THREADS ? (n1 < 64 ? 64 : n1) : NTT_THREADS;
         const uint32_t tr = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
-        fp256* tmp = ds.d_ntt_tmp;
-        // all DFT passes fully coalesced; tiled transposes carry the layout
-        hipLaunchKernelGGL(k_transpose, dim3(n2 / TT, n1 / TT), dim3(256), 0,
-                           st, d_data, tmp, n1, n2);
-        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(tc), LDS_BYTES(n1), st,
-                           tmp, plan->tw1, plan->twB, plan->twB + t1n,
-                           fwd_cT1, fwd_cT2, plan->k1, plan->k2);
-        hipLaunchKernelGGL(k_transpose, dim3(n1 / TT, n2 / TT), dim3(256), 0,
-                           st, tmp, d_data, n2, n1);
-        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(tr), LDS_BYTES(n2), st,
-                           d_data, tmp, plan->tw2, inv_cT1, inv_cT2,
+        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(tc),
+                           LDS_BYTES(n1), st, d_data,
+                           ds.d_ntt_tmp, plan->tw1, plan->twB,
+                           plan->twB + t1n, fwd_cT1, fwd_cT2, plan->k1,
+                           plan->k2);
+        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(tr),
+                           LDS_BYTES(n2), st, ds.d_ntt_tmp,
+                           d_data, plan->tw2, inv_cT1, inv_cT2,
                            /*coset_on_load=*/0, scale, inverse ? 1 : 0,
                            plan->k1, plan->k2);
-        hipLaunchKernelGGL(k_transpose, dim3(n2 / TT, n1 / TT), dim3(256), 0,
-                           st, tmp, d_data, n1, n2);
     } else {
         // single pass; forward coset applies on load, inverse on store
         const fp256* cc1 = coset_gen ? cT1 : nullptr;
