@@ -25,8 +25,8 @@
 #include "internal.hpp"
 
 #define THREADS 256       // pow-table builder
-#define NTT_THREADS 512   // 8 waves/block; with __launch_bounds__(512,2) the
-                          // register-DIF x[4] sets stay spill-free (256 VGPR)
+#define NTT_THREADS 1024  // NTT passes: 16 waves/block so one LDS-resident
+                          // block still puts 4 waves on every SIMD
 #define TW_LOW_BITS 12
 #define TW_LOW_MASK 0xfffu
 
@@ -72,63 +72,29 @@ __global__ void k_pow_table(fp256 base, fp256* __restrict__ out,
 
 // in-LDS DIF butterflies over L = 2^logL elements; twL[j] = (root)^j, j < L/2.
 // On exit lds[s] holds DFT output index bitrev(s, logL).
-// One group of G consecutive DIF stages (top stage half-size h = 1<<logh):
-// each thread loads a 2^G-element register set at stride S = h >> (G-1),
-// runs the G butterfly stages entirely in registers, stores back — ONE LDS
-// round trip + barrier instead of G. Identical butterfly network and
-// twiddle schedule to the plain radix-2 loop (output stays bit-reversed).
-template <int G>
-__device__ void dif_group(uint4* lds4, uint32_t H,
-                          const fp256* __restrict__ twL, uint32_t logL,
-                          uint32_t logh) {
-    constexpr uint32_t R = 1u << G;
-    const uint32_t L = 1u << logL;
-    const uint32_t logS = logh - (G - 1);
-    const uint32_t S = 1u << logS;
-    const uint32_t sets = L >> G;
-    for (uint32_t setid = threadIdx.x; setid < sets; setid += blockDim.x) {
-        const uint32_t base = ((setid >> logS) << (logS + G)) + (setid & (S - 1));
-        fp256 x[R];
-#pragma unroll
-        for (uint32_t m = 0; m < R; m++) lds_ld(lds4, H, base + m * S, x[m]);
-#pragma unroll
-        for (uint32_t k = 0; k < G; k++) {
-            const uint32_t hh = R >> (k + 1);       // local half-size
-            const uint32_t hk_mask = (1u << (logh - k)) - 1;  // global h - 1
-            const uint32_t stride = (L >> 1) >> (logh - k);
-#pragma unroll
-            for (uint32_t p = 0; p < (R >> 1); p++) {
-                const uint32_t blk = p / hh, mj = p % hh;
-                const uint32_t m0 = blk * 2 * hh + mj, m1 = m0 + hh;
-                const uint32_t j = (base + m0 * S) & hk_mask;
-                fp256 u = x[m0], v = x[m1];
-                ff_add<Fr>(x[m0], u, v);
-                ff_sub<Fr>(v, u, v);
-                ff_mul<Fr>(x[m1], v, twL[(uint64_t)j * stride]);
-            }
-        }
-#pragma unroll
-        for (uint32_t m = 0; m < R; m++) lds_st(lds4, H, base + m * S, x[m]);
-    }
-    __syncthreads();
-}
-
 __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
                         uint32_t logL) {
-    // G=2 groups (x[4] register sets): 6 LDS round trips for a 2^12 tile
-    // instead of 12 (G=3's x[8] spills even at the 256-VGPR budget).
-    int left = (int)logL;
-    uint32_t logh = logL - 1;
-    while (left >= 2) {
-        dif_group<2>(lds4, H, twL, logL, logh);
-        logh -= 2;
-        left -= 2;
+    const uint32_t L = 1u << logL;
+    for (uint32_t h = L >> 1; h >= 1; h >>= 1) {
+        const uint32_t stride = (L >> 1) / h;
+        for (uint32_t p = threadIdx.x; p < (L >> 1); p += blockDim.x) {
+            uint32_t blk = p / h, j = p % h;
+            uint32_t i0 = blk * 2 * h + j, i1 = i0 + h;
+            fp256 u, v, t, w;
+            lds_ld(lds4, H, i0, u);
+            lds_ld(lds4, H, i1, v);
+            ff_add<Fr>(w, u, v);
+            lds_st(lds4, H, i0, w);
+            ff_sub<Fr>(t, u, v);
+            ff_mul<Fr>(t, t, twL[(uint64_t)j * stride]);
+            lds_st(lds4, H, i1, t);
+        }
+        __syncthreads();
     }
-    if (left == 1) dif_group<1>(lds4, H, twL, logL, logh);
 }
 
 // pass A: column DFTs. grid.x = n2; LDS = n1 elements.
-__global__ __launch_bounds__(NTT_THREADS, 2) void k_ntt_col(
+__global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
                           const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw1,
                           const fp256* __restrict__ T1,
@@ -166,7 +132,7 @@ __global__ __launch_bounds__(NTT_THREADS, 2) void k_ntt_col(
 
 // pass B: row DFTs + transposed store. grid.x = n1; LDS = n2 elements.
 // in == out is safe only when n1 == 1 (single workgroup).
-__global__ __launch_bounds__(NTT_THREADS, 2) void k_ntt_row(
+__global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
                           const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw2,
                           const fp256* __restrict__ cT1,
