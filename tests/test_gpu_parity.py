@@ -253,3 +253,39 @@ def test_ntt_full_size_spotcheck_vs_dft(gpu, oracle, golden):
             cur = cur * wj % r
         got = from_mont(out[32 * j:32 * j + 32])
         assert got == acc, f"output {j}"
+
+
+# ---------------------------------------------------------------- robustness
+def test_concurrent_calls_one_ctx(gpu, oracle):
+    """halo2 commits from concurrent rayon workers; the ctx must serialize
+    safely and return correct results from every thread."""
+    import threading
+    n = 1024
+    cases = []
+    for seed in range(6):
+        sc, bs = oracle.gen_msm_inputs(n, 800 + seed, fast=True)
+        cases.append((sc, bs, oracle.msm(bs, sc, n)))
+    errors = []
+
+    def worker(sc, bs, want):
+        try:
+            for _ in range(3):
+                got = gpu.msm(bs, sc, n)
+                assert got == want
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=c) for c in cases]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors
+
+
+def test_error_paths(gpu):
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="log_n"):
+        gpu.ntt(b"\x00" * 32, 25, b"\x01" + b"\x00" * 31)
+    with _pytest.raises(RuntimeError, match="nbatch"):
+        gpu.msm_batch(b"\x00" * 64, b"\x00" * (33 * 32), 33, 1)
