@@ -349,6 +349,10 @@ int spectre_gpu_ntt_fr(spectre_gpu_ctx* ctx, uint8_t* data, uint32_t log_n,
                        const uint8_t omega[32], int inverse,
                        const uint8_t* coset_gen) {
     if (check_dev(ctx, 0)) return -1;
+    if (log_n > 24) {
+        set_err("ntt: log_n %u > 24 unsupported", log_n);
+        return -3;
+    }
     std::lock_guard<std::recursive_mutex> lk(ctx->mu);
     DeviceState& ds = ctx->devs[0];
     HIP_TRY(hipSetDevice(ds.device_id));
