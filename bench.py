@@ -55,10 +55,12 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     dist = None
+    backend = os.environ.get("SPECTRE_BENCH_BACKEND", "nccl")
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        dist.init_process_group("nccl")
+        dist.init_process_group(backend)
+        local_rank %= max(1, torch.cuda.device_count())
         torch.cuda.set_device(local_rank)
     n_gpus = world if world > 1 else args.gpus
     if world == 1 and args.gpus > 1:
@@ -85,7 +87,9 @@ def main():
     def step():
         partials = gpu.msm_shard_device(d_b, d_s, m)
         if dist is not None:
-            t = torch.frombuffer(bytearray(partials), dtype=torch.uint8).to(dev)
+            t = torch.frombuffer(bytearray(partials), dtype=torch.uint8)
+            if backend == "nccl":
+                t = t.to(dev)  # 1.5 KiB window sums over RCCL/xGMI
             gath = [torch.empty_like(t) for _ in range(world)]
             dist.all_gather(gath, t)
             blob = b"".join(g.cpu().numpy().tobytes() for g in gath)
