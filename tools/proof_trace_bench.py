@@ -1,0 +1,166 @@
+#!/usr/bin/env python3
+"""Replay of the sync-step k=20 `create_proof` hot-path call sequence
+(SURVEY.md §7 step 5 / §3a): the GPU-side wall-clock of every MSM and NTT a
+single sync-step proof issues through the best_multiexp/best_fft seam,
+against the same sequence on the CPU oracle (the ">=10x vs host CPU"
+check at proof granularity, not just per-kernel).
+
+Call counts are the structural estimates of SURVEY.md §3a for the pinned
+k=20 config (15 advice + 2 lookup-advice columns, lookup_bits 19, extended
+domain 2^22); exact counts await the instrumented Rust shim
+in a cargo-capable environment, which logs them for free. Witness data is
+synthetic (the arithmetic is shape-dependent only).
+
+Run on a GPU box:   python tools/proof_trace_bench.py [--cpu]
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+HERE = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, HERE)
+sys.path.insert(0, os.path.join(HERE, "oracle"))
+import pywrap as oracle  # noqa: E402  (CPU baseline leg + input gen only)
+
+K = 20
+N = 1 << K
+EXT_K = 22  # extended/quotient domain at k=20 (extended_k = k + 2)
+R = 21888242871839275222246405745257275088548364400416034343698204186575808495617
+
+# (phase, kind, log_n or n, count) — SURVEY.md §3a sync-step k=20 estimates
+TRACE = [
+    ("advice commit",      "msm",  K,     17),  # 15 advice + 2 lookup-advice
+    ("lookup commits",     "msm",  K,      6),  # ~3 per lookup argument x2
+    ("permutation z",      "msm",  K,      6),
+    ("quotient iFFT",      "intt", K,     30),  # per committed poly
+    ("quotient coset-FFT", "coset", EXT_K, 30),  # extended domain
+    ("quotient h commits", "msm",  EXT_K,  4),  # h(X) pieces (2^22 each)
+    ("coset iFFT",         "icoset", EXT_K, 1),
+    ("SHPLONK multiopen",  "msm",  K,      2),
+]
+
+
+def omega_for(log_n):
+    g7 = oracle.fr_from_canonical((7).to_bytes(32, "little"))
+    w = oracle.fr_pow(g7, ((R - 1) >> 28).to_bytes(32, "little"))
+    for _ in range(28 - log_n):
+        w = oracle.fr_mul(w, w)
+    return w
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--cpu", action="store_true",
+                    help="also time the CPU-oracle replay (slow)")
+    args = ap.parse_args()
+    from spectre_amd import SpectreGpu
+
+    gpu = SpectreGpu([0])
+    g5 = oracle.fr_from_canonical((5).to_bytes(32, "little"))
+    g5i = oracle.fr_inv(g5)
+
+    print("[trace] generating synthetic inputs ...", file=sys.stderr)
+    sc20, bs20 = oracle.gen_msm_inputs(N, 42, fast=True)
+    _, bs22 = oracle.gen_msm_inputs(1 << EXT_K, 43, fast=True)
+    sc22 = oracle.gen_fr_vector(1 << EXT_K, 44)  # Montgomery; fine as scalars
+    vec20 = oracle.gen_fr_vector(N, 45)
+    vec22 = oracle.gen_fr_vector(1 << EXT_K, 46)
+
+    # device-resident state: SRS bases cached once (as the ctx cache would),
+    # polynomial buffers resident as the on-device pipeline keeps them
+    d_b20 = gpu.malloc(64 * N)
+    gpu.upload(d_b20, bs20)
+    d_b22 = gpu.malloc(64 << EXT_K)
+    gpu.upload(d_b22, bs22)
+    d_s20 = gpu.malloc(32 * N)
+    gpu.upload(d_s20, sc20)
+    d_s22 = gpu.malloc(32 << EXT_K)
+    gpu.upload(d_s22, sc22[: 32 << EXT_K])
+    # batched-commit scalar arrays (the 17 columns of phase 1 etc.; identical
+    # vectors are timing-equivalent to distinct columns of the same size)
+    d_batch20 = gpu.malloc(32 * N * 17)
+    for b in range(17):
+        gpu.upload(d_batch20 + b * 32 * N, sc20)
+    d_batch22 = gpu.malloc((32 << EXT_K) * 4)
+    for b in range(4):
+        gpu.upload(d_batch22 + b * (32 << EXT_K), sc22[: 32 << EXT_K])
+    d_v20 = gpu.malloc(32 * N)
+    gpu.upload(d_v20, vec20)
+    d_v22 = gpu.malloc(32 << EXT_K)
+    gpu.upload(d_v22, vec22)
+    w20, w22 = omega_for(K), omega_for(EXT_K)
+    w20i, w22i = oracle.fr_inv(w20), oracle.fr_inv(w22)
+
+    def run_phase(kind, log_n, count):
+        if kind == "msm":
+            n = 1 << log_n
+            d_b = d_b20 if log_n == K else d_b22
+            d_batch = d_batch20 if log_n == K else d_batch22
+            if count > 1:  # columns share bases: one fused batch call
+                gpu.msm_batch_device(d_b, d_batch, count, n)
+            else:
+                gpu.msm_shard_device(d_b, d_s20 if log_n == K else d_s22, n)
+        else:
+            d_v = d_v20 if log_n == K else d_v22
+            for _ in range(count):
+                if kind == "intt":
+                    gpu.ntt_device(d_v, log_n, w20i if log_n == K else w22i,
+                                   inverse=True)
+                elif kind == "coset":
+                    gpu.ntt_device(d_v, log_n, w20 if log_n == K else w22,
+                                   coset_gen=g5)
+                elif kind == "icoset":
+                    gpu.ntt_device(d_v, log_n, w20i if log_n == K else w22i,
+                                   inverse=True, coset_gen=g5i)
+
+    # warm (plans, scratch)
+    for phase, kind, log_n, count in TRACE:
+        run_phase(kind, log_n, 1)
+
+    total = 0.0
+    rows = []
+    for phase, kind, log_n, count in TRACE:
+        t0 = time.time()
+        run_phase(kind, log_n, count)
+        dt = time.time() - t0
+        total += dt
+        rows.append((phase, kind, log_n, count, dt))
+    for phase, kind, log_n, count, dt in rows:
+        print(f"  {phase:22s} {kind:6s} 2^{log_n} x{count:3d}: {dt * 1e3:9.2f} ms")
+    print(f"GPU hot-path total for one sync-step k=20 proof: {total * 1e3:.1f} ms")
+
+    out = {"trace": "sync-step k=20 (SURVEY §3a estimates)",
+           "gpu_hotpath_ms": round(total * 1e3, 1)}
+
+    if args.cpu:
+        print("[trace] CPU-oracle replay (OpenMP) ...", file=sys.stderr)
+        t0 = time.time()
+        for phase, kind, log_n, count in TRACE:
+            n = 1 << log_n
+            for _ in range(count):
+                if kind == "msm":
+                    oracle.msm(bs20 if log_n == K else bs22,
+                               sc20 if log_n == K else sc22[:32 * n], n)
+                elif kind == "intt":
+                    oracle.ntt(vec20 if log_n == K else vec22, log_n,
+                               w20i if log_n == K else w22i, inverse=True)
+                elif kind == "coset":
+                    oracle.ntt(vec20 if log_n == K else vec22, log_n,
+                               w20 if log_n == K else w22, coset_gen=g5)
+                elif kind == "icoset":
+                    oracle.ntt(vec20 if log_n == K else vec22, log_n,
+                               w20i if log_n == K else w22i, inverse=True,
+                               coset_gen=g5i)
+        cpu_total = time.time() - t0
+        out["cpu_hotpath_ms"] = round(cpu_total * 1e3, 1)
+        out["cpu_cores"] = oracle.num_threads()
+        out["speedup"] = round(cpu_total / total, 1)
+        print(f"CPU hot-path total ({oracle.num_threads()} cores): "
+              f"{cpu_total * 1e3:.1f} ms  ->  GPU speedup {out['speedup']}x")
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
