@@ -115,8 +115,8 @@ int spectre_gpu_msm_g1_combine(const uint8_t* partials, uint32_t nshards,
  *   data <- DFT(data, omega)     [caller passes omega_inv to invert]
  *   if inverse               : data[i] *= n^{-1}
  *   if coset_gen && inverse  : data[i] *= coset_gen^i   (after; pass g^{-1})
- * log_n <= 24 (2-adicity headroom and scratch sizing; the reference needs
- * 2^20..2^24). */
+ * log_n <= 28 (= BN254 Fr's 2-adicity; the reference needs 2^20..2^24 for
+ * its circuits and 2^25..2^26 extended domains for aggregation). */
 int spectre_gpu_ntt_fr(spectre_gpu_ctx*, uint8_t* data, uint32_t log_n,
                        const uint8_t omega[32], int inverse,
                        const uint8_t* coset_gen);

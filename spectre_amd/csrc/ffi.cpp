@@ -73,6 +73,7 @@ void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
         for (auto& kv : ds.plans) {
             if (kv.second.tw1) (void)hipFree(kv.second.tw1);
             if (kv.second.tw2) (void)hipFree(kv.second.tw2);
+            if (kv.second.tw3) (void)hipFree(kv.second.tw3);
             if (kv.second.twB) (void)hipFree(kv.second.twB);
         }
         if (ds.stream) (void)hipStreamDestroy(ds.stream);
@@ -349,8 +350,8 @@ int spectre_gpu_ntt_fr(spectre_gpu_ctx* ctx, uint8_t* data, uint32_t log_n,
                        const uint8_t omega[32], int inverse,
                        const uint8_t* coset_gen) {
     if (check_dev(ctx, 0)) return -1;
-    if (log_n > 24) {
-        set_err("ntt: log_n %u > 24 unsupported", log_n);
+    if (log_n > 28) {
+        set_err("ntt: log_n %u > 28 unsupported (Fr 2-adicity)", log_n);
         return -3;
     }
     std::lock_guard<std::recursive_mutex> lk(ctx->mu);

@@ -28,10 +28,11 @@
 #define MSM_ACC_E 64                       // sorted entries per acc thread
 
 struct NttPlan {
-    fp256* tw1 = nullptr;  // (w^{n2})^j, j < n1/2
-    fp256* tw2 = nullptr;  // (w^{n1})^j, j < n2/2
-    fp256* twB = nullptr;  // w^c, c < n2 (inter-pass twiddle bases)
-    uint32_t k1 = 0, k2 = 0;
+    fp256* tw1 = nullptr;  // butterfly twiddles, axis 1
+    fp256* tw2 = nullptr;  // butterfly twiddles, axis 2
+    fp256* tw3 = nullptr;  // butterfly twiddles, axis 3 (3-pass only)
+    fp256* twB = nullptr;  // T1 || T2 omega-power lookup (w^e = T1[e&fff]*T2[e>>12])
+    uint32_t k1 = 0, k2 = 0, k3 = 0;
 };
 
 struct CachedBases {

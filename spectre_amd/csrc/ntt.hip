@@ -20,6 +20,13 @@
 //   g^e = T1[e & 0xfff] * T2[e >> 12]   (one mul + two cached loads)
 // built once per (omega, log_n) plan and cached on device.
 //
+// For log_n in [25, 28] (the aggregation circuits' extended domains reach
+// 2^25..2^26; Fr's 2-adicity caps at 2^28) a THREE-pass variant (k_ntt_axis)
+// decomposes n = A*B*C with each axis <= 2^12:
+//   X[tA + A*tB + A*B*tC] = DFT_C(twiddle(DFT_B(twiddle(DFT_A(x)))))
+// with post-twiddles w^(tA*(b*C+c)) after pass 1 and w^(A*tB*c) after pass 2
+// (derivation in the k_ntt_axis comment).
+//
 // All Fr math is 8x32-limb Montgomery (ff.hpp); data stays in Montgomery form
 // end-to-end exactly as halo2 holds its &[Fr] slices.
 #include "internal.hpp"
@@ -169,6 +176,71 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
     }
 }
 
+// generic strided-axis DFT for the three-pass (log_n > 24) path.
+// Transform m (grid.x) covers element j at
+//   pos(m, j) = (m >> logS) << (logL + logS) | (m & (S-1)) | j*S-positioned,
+// i.e. base + j*S with base = (m >> logS)*L*S + (m & (S-1)).
+// Derivation (i = a*BC + b*C + c, t = tA + A*tB + AB*tC):
+//   i*t mod n = a*BC*tA + b*C*(tA + A*tB) + c*(tA + A*tB + AB*tC)
+// so pass1 (axis a, root w^(BC)) is followed by twiddle w^(tA*(b*C+c)) =
+// w^(t*m); pass2 (axis b, root w^(AC)) by w^(A*tB*c) = w^((t<<logA)*(m&(C-1)));
+// pass3 (axis c, root w^(AB)) scatters to out[tA + A*tB + AB*t] and carries
+// the n^{-1} scale / inverse-coset factor. All exponents < n <= 2^28 (u32).
+__global__ __launch_bounds__(NTT_THREADS) void k_ntt_axis(
+    const fp256* __restrict__ in, fp256* __restrict__ out,
+    const fp256* __restrict__ twL, const fp256* __restrict__ T1,
+    const fp256* __restrict__ T2, const fp256* __restrict__ cT1,
+    const fp256* __restrict__ cT2, int coset_on_load, fp256 scale,
+    int apply_scale, uint32_t logL, uint32_t logS, uint32_t post_mode,
+    uint32_t post_logA, uint32_t post_logC, int final_scatter, uint32_t fin_A,
+    uint32_t fin_B) {
+    extern __shared__ uint4 lds4[];
+    const uint32_t L = 1u << logL;
+    const uint32_t S = 1u << logS;
+    const uint32_t H = L;
+    const uint32_t m = blockIdx.x;
+    const uint64_t base =
+        ((uint64_t)(m >> logS) << (logL + logS)) + (m & (S - 1));
+    for (uint32_t s = threadIdx.x; s < L; s += blockDim.x) {
+        const uint64_t pos = base + (uint64_t)s * S;
+        fp256 v = in[pos];
+        if (cT1 && coset_on_load) {  // forward coset: g^(original index)
+            fp256 f;
+            tw_lookup(f, cT1, cT2, (uint32_t)pos);
+            ff_mul<Fr>(v, v, f);
+        }
+        lds_st(lds4, H, s, v);
+    }
+    __syncthreads();
+    lds_dif(lds4, H, twL, logL);
+    for (uint32_t s = threadIdx.x; s < L; s += blockDim.x) {
+        const uint32_t t = bitrev(s, logL);
+        fp256 v, f;
+        lds_ld(lds4, H, s, v);
+        if (post_mode == 1) {
+            tw_lookup(f, T1, T2, t * m);
+            ff_mul<Fr>(v, v, f);
+        } else if (post_mode == 2) {
+            tw_lookup(f, T1, T2, (t << post_logA) * (m & ((1u << post_logC) - 1)));
+            ff_mul<Fr>(v, v, f);
+        }
+        if (apply_scale) ff_mul<Fr>(v, v, scale);
+        uint64_t opos;
+        if (final_scatter) {
+            const uint32_t tA = m / fin_B, tB = m % fin_B;
+            opos = (uint64_t)tA + (uint64_t)fin_A * tB +
+                   (uint64_t)fin_A * fin_B * t;
+            if (cT1 && !coset_on_load) {  // inverse coset: g^(output index)
+                tw_lookup(f, cT1, cT2, (uint32_t)opos);
+                ff_mul<Fr>(v, v, f);
+            }
+        } else {
+            opos = base + (uint64_t)t * S;
+        }
+        out[opos] = v;
+    }
+}
+
 // ---------------------------------------------------------------- host side
 static void host_pow_u32(fp256& o, const fp256& a, uint32_t e) {
     ff_pow_u32<Fr>(o, a, e);
@@ -193,24 +265,46 @@ static int get_plan(DeviceState& ds, const fp256& omega, uint32_t log_n,
         return 0;
     }
     NttPlan p;
-    // balanced split (both LDS tiles small -> more blocks/CU); k2 >= k1 so
-    // the contiguous row pass gets the bigger tile
-    p.k2 = log_n <= 12 ? log_n : (log_n + 1) / 2;
-    p.k1 = log_n - p.k2;
-    const uint32_t n1 = 1u << p.k1, n2 = 1u << p.k2;
+    const bool force3 = getenv("SPECTRE_NTT_FORCE3") != nullptr;
     const uint32_t t1n = (log_n < TW_LOW_BITS) ? (1u << log_n) : (1u << TW_LOW_BITS);
     const uint32_t t2n = (log_n > TW_LOW_BITS) ? (1u << (log_n - TW_LOW_BITS)) : 1;
-    if (p.k1) HIP_TRY(hipMalloc(&p.tw1, (n1 / 2) * sizeof(fp256)));
-    HIP_TRY(hipMalloc(&p.tw2, (n2 / 2 ? n2 / 2 : 1) * sizeof(fp256)));
-    HIP_TRY(hipMalloc(&p.twB, (t1n + t2n) * sizeof(fp256)));  // T1 || T2
-    fp256 w1, w2, wT2;
-    host_pow_u32(w1, omega, n2);                    // root of column DFT
-    host_pow_u32(w2, omega, n1);                    // root of row DFT
-    host_pow_u32(wT2, omega, 1u << TW_LOW_BITS);    // T2 base
-    if (p.k1) build_tables(ds, w1, p.tw1, n1 / 2);
-    build_tables(ds, w2, p.tw2, n2 / 2 ? n2 / 2 : 1);
+    HIP_TRY(hipMalloc(&p.twB, ((uint64_t)t1n + t2n) * sizeof(fp256)));  // T1||T2
+    fp256 wT2;
+    host_pow_u32(wT2, omega, 1u << TW_LOW_BITS);
     build_tables(ds, omega, p.twB, t1n);
     build_tables(ds, wT2, p.twB + t1n, t2n);
+    if (log_n > 24 || (force3 && log_n >= 3)) {
+        // three-pass split n = A*B*C, each axis <= 2^12
+        const uint32_t q = log_n / 3, r = log_n % 3;
+        p.k1 = q + (r > 0);  // kA
+        p.k2 = q + (r > 1);  // kB
+        p.k3 = q;            // kC
+        fp256 w1, w2, w3;
+        host_pow_u32(w1, omega, 1u << (p.k2 + p.k3));  // w^(B*C)
+        host_pow_u32(w2, omega, 1u << (p.k1 + p.k3));  // w^(A*C)
+        host_pow_u32(w3, omega, 1u << (p.k1 + p.k2));  // w^(A*B)
+        HIP_TRY(hipMalloc(&p.tw1, (1u << (p.k1 - 1)) * sizeof(fp256)));
+        HIP_TRY(hipMalloc(&p.tw2, (1u << (p.k2 - 1)) * sizeof(fp256)));
+        HIP_TRY(hipMalloc(&p.tw3, (1u << (p.k3 - 1)) * sizeof(fp256)));
+        build_tables(ds, w1, p.tw1, 1u << (p.k1 - 1));
+        build_tables(ds, w2, p.tw2, 1u << (p.k2 - 1));
+        build_tables(ds, w3, p.tw3, 1u << (p.k3 - 1));
+    } else {
+        // balanced 2-pass split; k2 >= k1 so the contiguous row pass gets
+        // the bigger tile
+        p.k2 = log_n <= 12 ? log_n : (log_n + 1) / 2;
+        p.k1 = log_n - p.k2;
+        const uint32_t n1 = 1u << p.k1, n2 = 1u << p.k2;
+        fp256 w1, w2;
+        host_pow_u32(w1, omega, n2);  // root of column DFT
+        host_pow_u32(w2, omega, n1);  // root of row DFT
+        if (p.k1) {
+            HIP_TRY(hipMalloc(&p.tw1, (n1 / 2) * sizeof(fp256)));
+            build_tables(ds, w1, p.tw1, n1 / 2);
+        }
+        HIP_TRY(hipMalloc(&p.tw2, (n2 / 2 ? n2 / 2 : 1) * sizeof(fp256)));
+        build_tables(ds, w2, p.tw2, n2 / 2 ? n2 / 2 : 1);
+    }
     HIP_TRY(hipStreamSynchronize(ds.stream));
     HIP_TRY(hipGetLastError());
     auto res = ds.plans.emplace(key, p);
@@ -222,8 +316,8 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
                const fp256& omega, int inverse, const fp256* coset_gen) {
     DeviceState& ds = ctx->devs[dev];
     HIP_TRY(hipSetDevice(ds.device_id));
-    if (log_n > 24) {
-        set_err("ntt: log_n %u > 24 unsupported", log_n);
+    if (log_n > 28) {  // BN254 Fr 2-adicity
+        set_err("ntt: log_n %u > 28 unsupported (Fr 2-adicity)", log_n);
         return -3;
     }
     if (log_n == 0) return 0;  // DFT of size 1 is the identity; n^{-1} = 1, g^0 = 1
@@ -271,7 +365,31 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
     const fp256* fwd_cT2 = (coset_gen && !inverse) ? cT2 : nullptr;
     const fp256* inv_cT1 = (coset_gen && inverse) ? cT1 : nullptr;
     const fp256* inv_cT2 = (coset_gen && inverse) ? cT2 : nullptr;
-    if (plan->k1 > 0) {
+    if (plan->k3 > 0) {
+        // three-pass path (log_n > 24, or SPECTRE_NTT_FORCE3 for testing)
+        const uint32_t kA = plan->k1, kB = plan->k2, kC = plan->k3;
+        const uint32_t A = 1u << kA, B = 1u << kB, C = 1u << kC;
+        fp256 one;
+        ff_set_one<Fr>(one);
+        const fp256* T1 = plan->twB;
+        const fp256* T2 = plan->twB + t1n;
+        auto tdim = [](uint32_t L) {
+            return L < NTT_THREADS ? (L < 64 ? 64u : L) : (uint32_t)NTT_THREADS;
+        };
+        hipLaunchKernelGGL(k_ntt_axis, dim3(B * C), dim3(tdim(A)),
+                           LDS_BYTES(A), st, d_data, ds.d_ntt_tmp, plan->tw1,
+                           T1, T2, fwd_cT1, fwd_cT2, /*coset_on_load=*/1, one,
+                           0, kA, kB + kC, /*post_mode=*/1, 0, 0, 0, 0, 0);
+        hipLaunchKernelGGL(k_ntt_axis, dim3(A * C), dim3(tdim(B)),
+                           LDS_BYTES(B), st, ds.d_ntt_tmp, ds.d_ntt_tmp,
+                           plan->tw2, T1, T2, nullptr, nullptr, 0, one, 0, kB,
+                           kC, /*post_mode=*/2, kA, kC, 0, 0, 0);
+        hipLaunchKernelGGL(k_ntt_axis, dim3(A * B), dim3(tdim(C)),
+                           LDS_BYTES(C), st, ds.d_ntt_tmp, d_data, plan->tw3,
+                           T1, T2, inv_cT1, inv_cT2, /*coset_on_load=*/0,
+                           scale, inverse ? 1 : 0, kC, 0, /*post_mode=*/0, 0,
+                           0, /*final_scatter=*/1, A, B);
+    } else if (plan->k1 > 0) {
         const uint32_t tc = n1 < NTT_THREADS ? (n1 < 64 ? 64 : n1) : NTT_THREADS;
         const uint32_t tr = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
         hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(tc),

@@ -193,7 +193,7 @@ class SpectreGpu:
 
     def ntt(self, data: bytes, log_n: int, omega: bytes, inverse: bool = False,
             coset_gen: bytes | None = None) -> bytes:
-        assert log_n > 24 or len(data) == 32 << log_n, "data length != 32*2^log_n"
+        assert log_n > 28 or len(data) == 32 << log_n, "data length != 32*2^log_n"
         buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
         om = (ctypes.c_uint8 * 32).from_buffer_copy(omega)
         cg = ((ctypes.c_uint8 * 32).from_buffer_copy(coset_gen)
