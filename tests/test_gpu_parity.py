@@ -297,6 +297,7 @@ def test_ntt_three_pass_forced_vs_oracle(oracle, golden):
     import os
     import subprocess
     import sys
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     code = r'''
 import sys, os
 sys.path.insert(0, %r); sys.path.insert(0, %r)
