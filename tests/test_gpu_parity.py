@@ -89,6 +89,16 @@ def test_msm_2pow20_vs_oracle(gpu, oracle):
     assert gpu.msm(bs, sc, n) == want
 
 
+@pytest.mark.parametrize("log_n", [22, 23])
+def test_msm_aggregation_sizes_vs_oracle(gpu, oracle, log_n):
+    """Aggregation-circuit MSM sizes (K=23/24 proofs commit at 2^23/2^24;
+    2^22 covers the k=20 extended-domain h commits)."""
+    n = 1 << log_n
+    sc, bs = oracle.gen_msm_inputs(n, 2200 + log_n, fast=True)
+    want = oracle.msm(bs, sc, n)
+    assert gpu.msm(bs, sc, n) == want
+
+
 def test_msm_bases_cache(gpu, oracle):
     n = 4096
     sc, bs = oracle.gen_msm_inputs(n, 31, fast=True)
