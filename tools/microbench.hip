@@ -80,5 +80,40 @@ int main() {
     double msm = time_kernel(k_mul_chain, pts, out, nwork, nwork / 256);
     printf("ff_mul chain: %.3f ms  (%.1f M mul/s)\n", msm,
            (double)nwork * ITERS * 16 / msm / 1e3);
+    // raw mad64 rate
+    {
+        hipEvent_t e0, e1;
+        hipEventCreate(&e0);
+        hipEventCreate(&e1);
+        uint32_t iters = 4096;
+        hipLaunchKernelGGL(k_mad64_rate, dim3(nwork / 256), dim3(256), 0, 0,
+                           (uint64_t*)out, iters);
+        hipDeviceSynchronize();
+        hipEventRecord(e0);
+        hipLaunchKernelGGL(k_mad64_rate, dim3(nwork / 256), dim3(256), 0, 0,
+                           (uint64_t*)out, iters);
+        hipEventRecord(e1);
+        hipEventSynchronize(e1);
+        float ms;
+        hipEventElapsedTime(&ms, e0, e1);
+        double mads = (double)nwork * iters * 4;
+        printf("v_mad_u64_u32 (4 indep chains): %.3f ms  %.2f T mad64/s "
+               "(peak-if-2cyc = 78.6T, 4cyc = 39.3T, 8cyc = 19.7T)\n", ms,
+               mads / ms / 1e9);
+    }
     return 0;
+}
+
+// raw v_mad_u64_u32 issue-rate probe: 4 independent mad chains per thread
+__global__ __launch_bounds__(256, 4) void k_mad64_rate(uint64_t* out,
+                                                       uint32_t iters) {
+    uint64_t a0 = threadIdx.x | 1, a1 = a0 + 3, a2 = a0 + 5, a3 = a0 + 7;
+    const uint32_t b = blockIdx.x | 3;
+    for (uint32_t i = 0; i < iters; i++) {
+        a0 = (uint64_t)(uint32_t)a0 * b + a1;
+        a1 = (uint64_t)(uint32_t)a1 * b + a2;
+        a2 = (uint64_t)(uint32_t)a2 * b + a3;
+        a3 = (uint64_t)(uint32_t)a3 * b + a0;
+    }
+    out[blockIdx.x * blockDim.x + threadIdx.x] = a0 + a1 + a2 + a3;
 }
