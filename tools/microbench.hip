@@ -10,6 +10,21 @@
 
 #define ITERS 256
 
+
+// raw v_mad_u64_u32 issue-rate probe: 4 independent mad chains per thread
+__global__ __launch_bounds__(256, 4) void k_mad64_rate(uint64_t* out,
+                                                       uint32_t iters) {
+    uint64_t a0 = threadIdx.x | 1, a1 = a0 + 3, a2 = a0 + 5, a3 = a0 + 7;
+    const uint32_t b = blockIdx.x | 3;
+    for (uint32_t i = 0; i < iters; i++) {
+        a0 = (uint64_t)(uint32_t)a0 * b + a1;
+        a1 = (uint64_t)(uint32_t)a1 * b + a2;
+        a2 = (uint64_t)(uint32_t)a2 * b + a3;
+        a3 = (uint64_t)(uint32_t)a3 * b + a0;
+    }
+    out[blockIdx.x * blockDim.x + threadIdx.x] = a0 + a1 + a2 + a3;
+}
+
 template <int CHAINS>
 __global__ __launch_bounds__(256, 2) void k_madd_chain(const g1_affine* pts,
                                                        g1_jac* out, int nwork) {
@@ -102,18 +117,4 @@ int main() {
                mads / ms / 1e9);
     }
     return 0;
-}
-
-// raw v_mad_u64_u32 issue-rate probe: 4 independent mad chains per thread
-__global__ __launch_bounds__(256, 4) void k_mad64_rate(uint64_t* out,
-                                                       uint32_t iters) {
-    uint64_t a0 = threadIdx.x | 1, a1 = a0 + 3, a2 = a0 + 5, a3 = a0 + 7;
-    const uint32_t b = blockIdx.x | 3;
-    for (uint32_t i = 0; i < iters; i++) {
-        a0 = (uint64_t)(uint32_t)a0 * b + a1;
-        a1 = (uint64_t)(uint32_t)a1 * b + a2;
-        a2 = (uint64_t)(uint32_t)a2 * b + a3;
-        a3 = (uint64_t)(uint32_t)a3 * b + a0;
-    }
-    out[blockIdx.x * blockDim.x + threadIdx.x] = a0 + a1 + a2 + a3;
 }
