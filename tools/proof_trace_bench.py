@@ -24,22 +24,45 @@ sys.path.insert(0, HERE)
 sys.path.insert(0, os.path.join(HERE, "oracle"))
 import pywrap as oracle  # noqa: E402  (CPU baseline leg + input gen only)
 
-K = 20
-N = 1 << K
-EXT_K = 22  # extended/quotient domain at k=20 (extended_k = k + 2)
 R = 21888242871839275222246405745257275088548364400416034343698204186575808495617
 
-# (phase, kind, log_n or n, count) — SURVEY.md §3a sync-step k=20 estimates
-TRACE = [
-    ("advice commit",      "msm",  K,     17),  # 15 advice + 2 lookup-advice
-    ("lookup commits",     "msm",  K,      6),  # ~3 per lookup argument x2
-    ("permutation z",      "msm",  K,      6),
-    ("quotient iFFT",      "intt", K,     30),  # per committed poly
-    ("quotient coset-FFT", "coset", EXT_K, 30),  # extended domain
-    ("quotient h commits", "msm",  EXT_K,  4),  # h(X) pieces (2^22 each)
-    ("coset iFFT",         "icoset", EXT_K, 1),
-    ("SHPLONK multiopen",  "msm",  K,      2),
-]
+# (phase, kind, log_n, count) — SURVEY.md §3a structural estimates.
+# step20: sync-step k=20 (15 advice + 2 lookup-advice cols, ext domain 2^22).
+# agg23: the sync-step-compressed aggregation stage (K=23, 1 advice col +
+#        lookup, ext domain 2^25). committee24: committee-update aggregation
+#        (K=24, ext 2^26).
+TRACES = {
+    "step20": (20, 22, [
+        ("advice commit",      "msm",   20, 17),
+        ("lookup commits",     "msm",   20,  6),
+        ("permutation z",      "msm",   20,  6),
+        ("quotient iFFT",      "intt",  20, 30),
+        ("quotient coset-FFT", "coset", 22, 30),
+        ("quotient h commits", "msm",   22,  4),
+        ("coset iFFT",         "icoset", 22, 1),
+        ("SHPLONK multiopen",  "msm",   20,  2),
+    ]),
+    "agg23": (23, 25, [
+        ("advice commit",      "msm",   23,  2),
+        ("lookup commits",     "msm",   23,  3),
+        ("permutation z",      "msm",   23,  2),
+        ("quotient iFFT",      "intt",  23,  8),
+        ("quotient coset-FFT", "coset", 25,  8),
+        ("quotient h commits", "msm",   23,  4),
+        ("coset iFFT",         "icoset", 25, 1),
+        ("SHPLONK multiopen",  "msm",   23,  2),
+    ]),
+    "committee24": (24, 26, [
+        ("advice commit",      "msm",   24,  2),
+        ("lookup commits",     "msm",   24,  3),
+        ("permutation z",      "msm",   24,  2),
+        ("quotient iFFT",      "intt",  24,  8),
+        ("quotient coset-FFT", "coset", 26,  8),
+        ("quotient h commits", "msm",   24,  4),
+        ("coset iFFT",         "icoset", 26, 1),
+        ("SHPLONK multiopen",  "msm",   24,  2),
+    ]),
+}
 
 
 def omega_for(log_n):
@@ -54,7 +77,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--cpu", action="store_true",
                     help="also time the CPU-oracle replay (slow)")
+    ap.add_argument("--config", default="step20", choices=sorted(TRACES))
     args = ap.parse_args()
+    global K, N, EXT_K, TRACE
+    K, EXT_K, TRACE = TRACES[args.config]
+    N = 1 << K
     from spectre_amd import SpectreGpu
 
     gpu = SpectreGpu([0])
@@ -63,7 +90,9 @@ def main():
 
     print("[trace] generating synthetic inputs ...", file=sys.stderr)
     sc20, bs20 = oracle.gen_msm_inputs(N, 42, fast=True)
-    _, bs22 = oracle.gen_msm_inputs(1 << EXT_K, 43, fast=True)
+    need_ext_msm = any(k == "msm" and ln != K for _, k, ln, c in TRACE)
+    bs22 = (oracle.gen_msm_inputs(1 << EXT_K, 43, fast=True)[1]
+            if need_ext_msm else b"")
     sc22 = oracle.gen_fr_vector(1 << EXT_K, 44)  # Montgomery; fine as scalars
     vec20 = oracle.gen_fr_vector(N, 45)
     vec22 = oracle.gen_fr_vector(1 << EXT_K, 46)
@@ -78,13 +107,17 @@ def main():
     gpu.upload(d_s20, sc20)
     d_s22 = gpu.malloc(32 << EXT_K)
     gpu.upload(d_s22, sc22[: 32 << EXT_K])
-    # batched-commit scalar arrays (the 17 columns of phase 1 etc.; identical
-    # vectors are timing-equivalent to distinct columns of the same size)
-    d_batch20 = gpu.malloc(32 * N * 17)
-    for b in range(17):
+    # batched-commit scalar arrays (the columns of each commit phase;
+    # identical vectors are timing-equivalent to distinct columns)
+    max20 = max((c for _, k, ln, c in TRACE if k == "msm" and ln == K),
+                default=1)
+    max22 = max((c for _, k, ln, c in TRACE if k == "msm" and ln != K),
+                default=1)
+    d_batch20 = gpu.malloc(32 * N * max20)
+    for b in range(max20):
         gpu.upload(d_batch20 + b * 32 * N, sc20)
-    d_batch22 = gpu.malloc((32 << EXT_K) * 4)
-    for b in range(4):
+    d_batch22 = gpu.malloc((32 << EXT_K) * max22)
+    for b in range(max22):
         gpu.upload(d_batch22 + b * (32 << EXT_K), sc22[: 32 << EXT_K])
     d_v20 = gpu.malloc(32 * N)
     gpu.upload(d_v20, vec20)
@@ -129,9 +162,9 @@ def main():
         rows.append((phase, kind, log_n, count, dt))
     for phase, kind, log_n, count, dt in rows:
         print(f"  {phase:22s} {kind:6s} 2^{log_n} x{count:3d}: {dt * 1e3:9.2f} ms")
-    print(f"GPU hot-path total for one sync-step k=20 proof: {total * 1e3:.1f} ms")
+    print(f"GPU hot-path total for one {args.config} proof: {total * 1e3:.1f} ms")
 
-    out = {"trace": "sync-step k=20 (SURVEY §3a estimates)",
+    out = {"trace": f"{args.config} (SURVEY §3a estimates)",
            "gpu_hotpath_ms": round(total * 1e3, 1)}
 
     if args.cpu:
