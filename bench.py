@@ -191,7 +191,7 @@ def main():
                 "workload": "msm_g1_2pow20",
                 "n": N,
                 "seed": SEED,
-                "parallelism": f"shard{n_gpus}" + ("+rccl" if world > 1 else ""),
+                "parallelism": f"shard{n_gpus}" + (f"+{backend}" if world > 1 else ""),
                 "scalars": "canonical",
             },
             "roofline": roofline,

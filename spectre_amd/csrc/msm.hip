@@ -237,9 +237,10 @@ PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
 // threads grid-stride their share, then a pairwise LDS tree — serial depth
 // ~16 adds in ONE launch (a 3-level cascade of tiny grids was ~0.8 ms of
 // launch+latency overhead).
-PT_KERNEL void k_window_sum(const g1_jac* __restrict__ in,
-                            g1_jac* __restrict__ out) {
-    __shared__ g1_jac lds[128];
+#define WSUM_THREADS 1024
+__global__ __launch_bounds__(WSUM_THREADS, 1) void k_window_sum(
+    const g1_jac* __restrict__ in, g1_jac* __restrict__ out) {
+    __shared__ g1_jac lds[WSUM_THREADS / 2];
     const uint32_t per_win = MSM_BPW / MSM_CHUNK;
     const uint32_t w = blockIdx.x;
     const uint32_t t = threadIdx.x;
@@ -247,7 +248,7 @@ PT_KERNEL void k_window_sum(const g1_jac* __restrict__ in,
     g1j_set_inf(acc);
     for (uint32_t j = t; j < per_win; j += blockDim.x)
         g1j_add_ip(acc, in[(uint64_t)w * per_win + j]);
-    for (uint32_t k = 128; k >= 1; k >>= 1) {
+    for (uint32_t k = WSUM_THREADS / 2; k >= 1; k >>= 1) {
         if (t >= k && t < 2 * k) lds[t - k] = acc;
         __syncthreads();
         if (t < k) g1j_add_ip(acc, lds[t]);
@@ -368,8 +369,8 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                        dim3((nchunks + THREADS - 1) / THREADS), dim3(THREADS),
                        0, st, ds.d_buckets, nchunks, red0);
     STAMP(5);
-    hipLaunchKernelGGL(k_window_sum, dim3(nbatch * MSM_NWIN), dim3(THREADS),
-                       0, st, red0, red1);
+    hipLaunchKernelGGL(k_window_sum, dim3(nbatch * MSM_NWIN),
+                       dim3(WSUM_THREADS), 0, st, red0, red1);
     STAMP(6);
     HIP_TRY(hipMemcpyAsync(winsums_host, red1,
                            (size_t)nbatch * MSM_NWIN * sizeof(g1_jac),
