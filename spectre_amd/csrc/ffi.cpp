@@ -325,9 +325,15 @@ int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
                                    hipMemcpyHostToDevice, ds.stream));
             d_b = ds.d_bases;
         }
-        int rc = msm_shard_device(ctx, d, d_b, ds.d_scalars, m, flags,
-                                  &wins[(size_t)d * MSM_NWIN]);
+        // enqueue without synchronizing so all shards run concurrently
+        int rc = msm_batch_shard_device(ctx, d, d_b, ds.d_scalars, 1, m,
+                                        flags, &wins[(size_t)d * MSM_NWIN],
+                                        nullptr, /*sync=*/false);
         if (rc) return rc;
+    }
+    for (int d = 0; d < num_gpus; d++) {
+        HIP_TRY(hipSetDevice(ctx->devs[d].device_id));
+        HIP_TRY(hipStreamSynchronize(ctx->devs[d].stream));
     }
     return spectre_gpu_msm_g1_combine((const uint8_t*)wins.data(),
                                       (uint32_t)num_gpus, out_affine);

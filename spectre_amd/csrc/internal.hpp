@@ -100,10 +100,14 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
                      g1_jac* winsums_host, double* stage_ms = nullptr);
 // batch variant: nbatch scalar vectors (batch-major, nbatch*n*32 B) over one
 // shared base set; winsums_host receives nbatch*MSM_NWIN Jacobian sums.
+// sync=false: enqueue the pipeline + async D2H of winsums on the device's
+// stream and return without synchronizing (multi-device overlap; the caller
+// must hipStreamSynchronize each stream before reading winsums_host).
 int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            const g1_affine* d_bases, const uint8_t* d_scalars,
                            uint32_t nbatch, uint64_t n, uint32_t flags,
-                           g1_jac* winsums_host, double* stage_ms = nullptr);
+                           g1_jac* winsums_host, double* stage_ms = nullptr,
+                           bool sync = true);
 
 // ntt.hip — in-place NTT on a device buffer (synchronizes the stream).
 int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,

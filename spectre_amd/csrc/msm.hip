@@ -310,7 +310,7 @@ static int end_bit_for(uint64_t max_key) {
 int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            const g1_affine* d_bases, const uint8_t* d_scalars,
                            uint32_t nbatch, uint64_t n, uint32_t flags,
-                           g1_jac* winsums_host, double* stage_ms) {
+                           g1_jac* winsums_host, double* stage_ms, bool sync) {
     DeviceState& ds = ctx->devs[dev];
     HIP_TRY(hipSetDevice(ds.device_id));
     if (nbatch == 0 || nbatch > SPECTRE_MSM_MAX_BATCH) {
@@ -379,6 +379,10 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
     if (stage_ms)
         HIP_TRY(hipMemcpyAsync(&ent_real, ds.d_offsets + nbt, 4,
                                hipMemcpyDeviceToHost, st));
+    if (!sync && !stage_ms) {
+        HIP_TRY(hipGetLastError());
+        return 0;  // caller synchronizes the stream before reading winsums
+    }
     HIP_TRY(hipStreamSynchronize(st));
     HIP_TRY(hipGetLastError());
     if (stage_ms) {
