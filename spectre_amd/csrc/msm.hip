@@ -146,7 +146,18 @@ PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
         }
         if (kj < nb_total) {
             uint32_t v = vals[j];
+#ifdef MSM_NT_LOADS
+            g1_affine p;
+            {
+                const g1_affine* src = &bases[v & 0x7fffffffu];
+                const uint32_t* s32 = (const uint32_t*)src;
+                uint32_t* d32 = (uint32_t*)&p;
+                for (int q = 0; q < 16; q++)
+                    d32[q] = __builtin_nontemporal_load(s32 + q);
+            }
+#else
             g1_affine p = bases[v & 0x7fffffffu];
+#endif
             if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
             g1j_madd_ip(acc, p);
         }
