@@ -68,6 +68,10 @@ void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
               (void*)ds.d_red, (void*)ds.d_scalars,
               (void*)ds.d_bases, (void*)ds.d_ntt_tmp, (void*)ds.d_cosetA})
             if (p) (void)hipFree(p);
+        for (int i = 0; i < 2; i++) {
+            if (ds.h_stage[i]) (void)hipHostFree(ds.h_stage[i]);
+            if (ds.stage_ev[i]) (void)hipEventDestroy(ds.stage_ev[i]);
+        }
         for (auto& kv : ds.bases_cache)
             if (kv.second.d_ptr) (void)hipFree(kv.second.d_ptr);
         for (auto& kv : ds.plans) {
@@ -83,6 +87,77 @@ void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
 
 int spectre_gpu_device_count(spectre_gpu_ctx* ctx) {
     return ctx ? (int)ctx->devs.size() : 0;
+}
+
+// ---------------------------------------------------------------- staging
+#define STAGE_CHUNK (16u << 20)
+
+static int ensure_stage(DeviceState& ds) {
+    if (!ds.h_stage[0]) {
+        HIP_TRY(hipHostMalloc((void**)&ds.h_stage[0], STAGE_CHUNK));
+        HIP_TRY(hipHostMalloc((void**)&ds.h_stage[1], STAGE_CHUNK));
+        HIP_TRY(hipEventCreate(&ds.stage_ev[0]));
+        HIP_TRY(hipEventCreate(&ds.stage_ev[1]));
+        HIP_TRY(hipEventRecord(ds.stage_ev[0], ds.stream));
+        HIP_TRY(hipEventRecord(ds.stage_ev[1], ds.stream));
+    }
+    return 0;
+}
+
+// host->device through alternating pinned chunks: memcpy of chunk k+1
+// overlaps the DMA of chunk k; the stream stays ordered so following
+// kernels need no extra sync. The user buffer is fully consumed by the
+// memcpys before return (the in-flight DMA reads only pinned memory).
+// double-buffered pinned-staging copies on ds.stream: pageable hipMemcpy
+// runs at ~3 GB/s; staging through pinned chunks overlaps the host memcpy
+// with the DMA and reaches ~3-4x that.
+static int staged_upload(DeviceState& ds, void* d_dst, const void* src,
+                         size_t bytes) {
+    int rc = ensure_stage(ds);
+    if (rc) return rc;
+    size_t off = 0;
+    int b = 0;
+    while (off < bytes) {
+        const size_t len = bytes - off < STAGE_CHUNK ? bytes - off : STAGE_CHUNK;
+        HIP_TRY(hipEventSynchronize(ds.stage_ev[b]));  // buffer free?
+        memcpy(ds.h_stage[b], (const uint8_t*)src + off, len);
+        HIP_TRY(hipMemcpyAsync((uint8_t*)d_dst + off, ds.h_stage[b], len,
+                               hipMemcpyHostToDevice, ds.stream));
+        HIP_TRY(hipEventRecord(ds.stage_ev[b], ds.stream));
+        off += len;
+        b ^= 1;
+    }
+    return 0;
+}
+
+static int staged_download(DeviceState& ds, void* dst, const void* d_src,
+                           size_t bytes) {
+    int rc = ensure_stage(ds);
+    if (rc) return rc;
+    size_t off = 0;
+    int b = 0;
+    size_t pend_off[2] = {0, 0}, pend_len[2] = {0, 0};
+    while (off < bytes) {
+        const size_t len = bytes - off < STAGE_CHUNK ? bytes - off : STAGE_CHUNK;
+        if (pend_len[b]) {  // drain the older use of this buffer
+            HIP_TRY(hipEventSynchronize(ds.stage_ev[b]));
+            memcpy((uint8_t*)dst + pend_off[b], ds.h_stage[b], pend_len[b]);
+        }
+        HIP_TRY(hipMemcpyAsync(ds.h_stage[b], (const uint8_t*)d_src + off, len,
+                               hipMemcpyDeviceToHost, ds.stream));
+        HIP_TRY(hipEventRecord(ds.stage_ev[b], ds.stream));
+        pend_off[b] = off;
+        pend_len[b] = len;
+        off += len;
+        b ^= 1;
+    }
+    for (int i = 0; i < 2; i++) {
+        if (pend_len[i]) {
+            HIP_TRY(hipEventSynchronize(ds.stage_ev[i]));
+            memcpy((uint8_t*)dst + pend_off[i], ds.h_stage[i], pend_len[i]);
+        }
+    }
+    return 0;
 }
 
 // ---------------------------------------------------------------- helpers
@@ -214,8 +289,10 @@ int spectre_gpu_msm_g1_batch(spectre_gpu_ctx* ctx, uint64_t bases_id,
         HIP_TRY(hipMalloc(&ds.d_scalars, sbytes));
         ds.scal_cap = sbytes;
     }
-    HIP_TRY(hipMemcpyAsync(ds.d_scalars, scalars, sbytes,
-                           hipMemcpyHostToDevice, ds.stream));
+    {
+        int rc = staged_upload(ds, ds.d_scalars, scalars, sbytes);
+        if (rc) return rc;
+    }
     g1_affine* d_b = nullptr;
     if (bases_id != 0) {
         auto key = std::make_pair(bases_id, n * 16 + 1);
@@ -231,8 +308,8 @@ int spectre_gpu_msm_g1_batch(spectre_gpu_ctx* ctx, uint64_t bases_id,
             CachedBases cb;
             cb.n = n;
             HIP_TRY(hipMalloc(&cb.d_ptr, n * sizeof(g1_affine)));
-            HIP_TRY(hipMemcpyAsync(cb.d_ptr, bases, n * 64,
-                                   hipMemcpyHostToDevice, ds.stream));
+            int rc = staged_upload(ds, cb.d_ptr, bases, n * 64);
+            if (rc) return rc;
             ds.bases_cache.emplace(key, cb);
             d_b = cb.d_ptr;
         }
@@ -246,8 +323,8 @@ int spectre_gpu_msm_g1_batch(spectre_gpu_ctx* ctx, uint64_t bases_id,
             HIP_TRY(hipMalloc(&ds.d_bases, n * sizeof(g1_affine)));
             ds.base_cap = n;
         }
-        HIP_TRY(hipMemcpyAsync(ds.d_bases, bases, n * 64,
-                               hipMemcpyHostToDevice, ds.stream));
+        int rc = staged_upload(ds, ds.d_bases, bases, n * 64);
+        if (rc) return rc;
         d_b = ds.d_bases;
     }
     return spectre_gpu_msm_g1_batch_device(ctx, 0, d_b, ds.d_scalars, nbatch,
@@ -287,8 +364,10 @@ int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
             HIP_TRY(hipMalloc(&ds.d_scalars, m * 32));
             ds.scal_cap = m * 32;
         }
-        HIP_TRY(hipMemcpyAsync(ds.d_scalars, scalars + lo * 32, m * 32,
-                               hipMemcpyHostToDevice, ds.stream));
+        {
+            int rc = staged_upload(ds, ds.d_scalars, scalars + lo * 32, m * 32);
+            if (rc) return rc;
+        }
         // bases: cached per (bases_id, n, num_gpus) — shard layout depends on
         // all three, so they are all folded into the cache key.
         g1_affine* d_b = nullptr;
@@ -306,8 +385,8 @@ int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
                 CachedBases cb;
                 cb.n = m;
                 HIP_TRY(hipMalloc(&cb.d_ptr, m * sizeof(g1_affine)));
-                HIP_TRY(hipMemcpyAsync(cb.d_ptr, bases + lo * 64, m * 64,
-                                       hipMemcpyHostToDevice, ds.stream));
+                int rc = staged_upload(ds, cb.d_ptr, bases + lo * 64, m * 64);
+                if (rc) return rc;
                 ds.bases_cache.emplace(key, cb);
                 d_b = cb.d_ptr;
             }
@@ -321,8 +400,8 @@ int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
                 HIP_TRY(hipMalloc(&ds.d_bases, m * sizeof(g1_affine)));
                 ds.base_cap = m;
             }
-            HIP_TRY(hipMemcpyAsync(ds.d_bases, bases + lo * 64, m * 64,
-                                   hipMemcpyHostToDevice, ds.stream));
+            int rc = staged_upload(ds, ds.d_bases, bases + lo * 64, m * 64);
+            if (rc) return rc;
             d_b = ds.d_bases;
         }
         // enqueue without synchronizing so all shards run concurrently
@@ -366,16 +445,11 @@ int spectre_gpu_ntt_fr(spectre_gpu_ctx* ctx, uint8_t* data, uint32_t log_n,
     const uint64_t n = 1ull << log_n;
     void* d_buf = nullptr;
     HIP_TRY(hipMalloc(&d_buf, n * 32));
-    HIP_TRY(hipMemcpyAsync(d_buf, data, n * 32, hipMemcpyHostToDevice,
-                           ds.stream));
-    int rc = spectre_gpu_ntt_fr_device(ctx, 0, d_buf, log_n, omega, inverse,
+    int rc = staged_upload(ds, d_buf, data, n * 32);
+    if (rc == 0)
+        rc = spectre_gpu_ntt_fr_device(ctx, 0, d_buf, log_n, omega, inverse,
                                        coset_gen);
-    if (rc == 0) {
-        rc = (hipMemcpy(data, d_buf, n * 32, hipMemcpyDeviceToHost) ==
-              hipSuccess)
-                 ? 0
-                 : -2;
-    }
+    if (rc == 0) rc = staged_download(ds, data, d_buf, n * 32);
     (void)hipFree(d_buf);
     return rc;
 }

@@ -64,6 +64,9 @@ struct DeviceState {
     g1_affine* d_bases = nullptr;
     size_t base_cap = 0;  // points
     std::map<std::pair<uint64_t, uint64_t>, CachedBases> bases_cache;  // (id, n)
+    // ---- pinned staging for host-pointer uploads/downloads ----
+    uint8_t* h_stage[2] = {nullptr, nullptr};
+    hipEvent_t stage_ev[2] = {nullptr, nullptr};
     // ---- NTT scratch ----
     fp256* d_ntt_tmp = nullptr;
     size_t ntt_cap = 0;  // elements

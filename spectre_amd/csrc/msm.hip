@@ -146,18 +146,9 @@ PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
         }
         if (kj < nb_total) {
             uint32_t v = vals[j];
-#ifdef MSM_NT_LOADS
-            g1_affine p;
-            {
-                const g1_affine* src = &bases[v & 0x7fffffffu];
-                const uint32_t* s32 = (const uint32_t*)src;
-                uint32_t* d32 = (uint32_t*)&p;
-                for (int q = 0; q < 16; q++)
-                    d32[q] = __builtin_nontemporal_load(s32 + q);
-            }
-#else
+            // plain (cached) loads: bases are re-read by all 16 windows, so
+            // L1/L2 residency pays (non-temporal loads measured 5% slower)
             g1_affine p = bases[v & 0x7fffffffu];
-#endif
             if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
             g1j_madd_ip(acc, p);
         }
