@@ -156,8 +156,9 @@ class SpectreGpu:
             canonical: bool = True, num_gpus: int = 1,
             bases_id: int = 0) -> bytes:
         out = (ctypes.c_uint8 * 64)()
-        b = (ctypes.c_uint8 * len(bases)).from_buffer_copy(bases) if bases else None
-        s = (ctypes.c_uint8 * len(scalars)).from_buffer_copy(scalars)
+        # read-only args pass zero-copy (the C side never writes them)
+        b = ctypes.cast(ctypes.c_char_p(bases), ctypes.c_void_p) if bases else None
+        s = ctypes.cast(ctypes.c_char_p(scalars), ctypes.c_void_p)
         rc = self._lib.spectre_gpu_msm_g1(
             self._ctx, bases_id, b, s, n,
             SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, num_gpus,
@@ -170,8 +171,8 @@ class SpectreGpu:
         """nbatch MSMs over one shared base set (batch-major scalars)."""
         assert len(scalars) == nbatch * n * 32
         out = (ctypes.c_uint8 * (64 * nbatch))()
-        b = (ctypes.c_uint8 * len(bases)).from_buffer_copy(bases) if bases else None
-        s = (ctypes.c_uint8 * len(scalars)).from_buffer_copy(scalars)
+        b = ctypes.cast(ctypes.c_char_p(bases), ctypes.c_void_p) if bases else None
+        s = ctypes.cast(ctypes.c_char_p(scalars), ctypes.c_void_p)
         rc = self._lib.spectre_gpu_msm_g1_batch(
             self._ctx, bases_id, b, s, nbatch, n,
             SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out)
@@ -263,7 +264,7 @@ class SpectreGpu:
                                                ctypes.c_void_p(d_ptr)), "free")
 
     def upload(self, d_dst: int, data: bytes, dev: int = 0) -> None:
-        buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
+        buf = ctypes.cast(ctypes.c_char_p(data), ctypes.c_void_p)
         self._check(self._lib.spectre_gpu_upload(self._ctx, dev,
                                                  ctypes.c_void_p(d_dst), buf,
                                                  len(data)), "upload")
