@@ -30,7 +30,7 @@
 #define MSM_BPW (1u << (MSM_WBITS - 1))    // buckets per window
 #define MSM_NB_TOTAL (MSM_NWIN * MSM_BPW)  // 262144
 #define MSM_SKIP_KEY MSM_NB_TOTAL          // sorts after all real keys
-#define MSM_CHUNK 4                        // buckets per reduction thread
+#define MSM_CHUNK 8                        // buckets per reduction thread
 #define MSM_ACC_E 64                       // sorted entries per acc thread
 
 struct NttPlan {
