@@ -42,7 +42,7 @@ typedef struct spectre_gpu_ctx spectre_gpu_ctx;
 #define SPECTRE_SCALARS_CANONICAL 1u  /* already to_repr()'d */
 
 #define SPECTRE_MSM_WINDOW_BITS 16u /* signed window width c */
-#define SPECTRE_MSM_NUM_WINDOWS 8u  /* GLV: ceil(128/c) windows per half */
+#define SPECTRE_MSM_NUM_WINDOWS 16u /* ceil(255/c) windows */
 #define SPECTRE_MSM_MAX_BATCH 32    /* max scalar vectors per batched call */
 
 /* ctx over `device_count` HIP devices (device_ids NULL -> 0..count-1).

@@ -18,18 +18,12 @@
 // and never carries out. Measured on MI355X: c=15 trades +8% bucket work
 // for a 2x smaller tail and nets out slightly WORSE than c=16 (5.14 vs
 // 5.02 ms at n=2^20) - c=16 kept.
-// GLV endomorphism (BN254: phi(x,y) = (beta*x, y) = lambda*P): every scalar
-// splits as k = k1 + lambda*k2 with |k1|,|k2| < 2^127 (constants + bound
-// validation in msm.hip), so the window set covers 128 bits instead of 255:
-// same total bucket-add work (2 halves x 8 windows = 16 entries/point) but
-// HALF the buckets -> half the reduction-tail work.
 #define MSM_WBITS 16
-#define MSM_NWIN 8                         // ceil(128 / WBITS) per GLV half
-#define MSM_HALVES 2
-#define MSM_ENT_PER (MSM_HALVES * MSM_NWIN)  // sort entries per point
+#define MSM_NWIN 16                        // ceil(255 / WBITS)
 #define MSM_BPW (1u << (MSM_WBITS - 1))    // buckets per window
-#define MSM_NB_TOTAL (MSM_NWIN * MSM_BPW)  // 262144
+#define MSM_NB_TOTAL (MSM_NWIN * MSM_BPW)
 #define MSM_SKIP_KEY MSM_NB_TOTAL          // sorts after all real keys
+#define MSM_SORT_BITS 20                   // key range < 2^20
 #define MSM_CHUNK 8                        // buckets per reduction thread
 #define MSM_ACC_E 64                       // sorted entries per acc thread
 

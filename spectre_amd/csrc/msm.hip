@@ -45,95 +45,7 @@
 // (measured: 1 add-chain/thread already saturates the VALU issue pipe).
 #define PT_KERNEL __global__ __launch_bounds__(THREADS, 2)
 
-// ---- GLV constants (derivation + 50k-sample bound validation in the
-// repo's round-1 notes; c1=(k*G1C)>>384 < 2^64, c2=(k*G2C)>>384 < 2^127,
-// k1 = k - c1*A1 - c2*A2, k2 = c1*|B1| - c2*B2, |k1|,|k2| < 2^127) --------
-static __device__ __constant__ uint32_t GLV_G1C[7] = {0x2fafba64u, 0x8fa7d32du, 0x773a6ef2u, 0x6eb9c714u, 0xc7e0b3d7u, 0xd91d232eu, 0x00000002u};
-static __device__ __constant__ uint32_t GLV_G2C[9] = {0x9b9bdffau, 0x86937516u, 0x5eaa26d9u, 0xa5e38cfbu, 0x391eb18du, 0x7a7bd9d4u, 0xa773d2cfu, 0x4ccef014u, 0x00000002u};
-static __device__ __constant__ uint32_t GLV_A1[2] = {0x94d213e3u, 0x89d32568u};
-static __device__ __constant__ uint32_t GLV_A2[4] = {0x1221250bu, 0x0be4e154u, 0xeeb859fdu, 0x6f4d8248u};
-static __device__ __constant__ uint32_t GLV_B1ABS[4] = {0x7d4f1128u, 0x8211bbebu, 0xeeb859fcu, 0x6f4d8248u};
-static __device__ __constant__ uint32_t GLV_B2[2] = {0x94d213e3u, 0x89d32568u};
-static __device__ __constant__ uint32_t GLV_BETA_MONT[8] = {0xd782e155u, 0x71930c11u, 0xffbe3323u, 0xa6bb947cu, 0xd4741444u, 0xaa303344u, 0x26594943u, 0x2c3b3f0du};
-
-// schoolbook a(na) * b(nb) -> out(na+nb), small fixed sizes
-template <int NA, int NB>
-__device__ __forceinline__ void mul_wide(const uint32_t* a, const uint32_t* b,
-                                         uint32_t* out) {
-    for (int i = 0; i < NA + NB; i++) out[i] = 0;
-    for (int i = 0; i < NA; i++) {
-        uint64_t cc = 0;
-        for (int j = 0; j < NB; j++) {
-            uint64_t x = (uint64_t)a[i] * b[j] + out[i + j] + (uint32_t)cc;
-            out[i + j] = (uint32_t)x;
-            cc = x >> 32;
-        }
-        uint64_t x = (uint64_t)out[i + NB] + (uint32_t)cc;
-        out[i + NB] = (uint32_t)x;
-        for (int j = i + NB + 1; (x >> 32) && j < NA + NB; j++) {
-            x = (uint64_t)out[j] + 1u;
-            out[j] = (uint32_t)x;
-        }
-    }
-}
-// out -= b over N limbs (two's complement), returns borrow
-template <int N>
-__device__ __forceinline__ uint32_t sub_n(uint32_t* out, const uint32_t* b) {
-    uint64_t brw = 0;
-    for (int i = 0; i < N; i++) {
-        uint64_t d = (uint64_t)out[i] - b[i] - brw;
-        out[i] = (uint32_t)d;
-        brw = (d >> 32) & 1;
-    }
-    return (uint32_t)brw;
-}
-template <int N>
-__device__ __forceinline__ void neg_n(uint32_t* out) {
-    uint64_t c = 1;
-    for (int i = 0; i < N; i++) {
-        c += (uint64_t)(~out[i]);
-        out[i] = (uint32_t)c;
-        c >>= 32;
-    }
-}
-
-// split canonical k into (|k1|, sign1, |k2|, sign2), magnitudes 4 limbs
-__device__ __forceinline__ void glv_decompose(const fp256& k, uint32_t m1[4],
-                                              uint32_t& neg1, uint32_t m2[4],
-                                              uint32_t& neg2) {
-    uint32_t prod[17];
-    mul_wide<8, 7>(k.l, GLV_G1C, prod);
-    uint32_t c1[2] = {prod[12], prod[13]};
-    mul_wide<8, 9>(k.l, GLV_G2C, prod);
-    uint32_t c2[4] = {prod[12], prod[13], prod[14], prod[15]};
-    // k1 = k - c1*A1 - c2*A2 in 256-bit two's complement
-    uint32_t k1[8];
-    for (int i = 0; i < 8; i++) k1[i] = k.l[i];
-    uint32_t t4[4], t8[8];
-    mul_wide<2, 2>(c1, GLV_A1, t4);
-    uint32_t t4e[8] = {t4[0], t4[1], t4[2], t4[3], 0, 0, 0, 0};
-    sub_n<8>(k1, t4e);
-    mul_wide<4, 4>(c2, GLV_A2, t8);
-    sub_n<8>(k1, t8);
-    neg1 = k1[7] >> 31;
-    if (neg1) neg_n<8>(k1);
-    for (int i = 0; i < 4; i++) m1[i] = k1[i];  // |k1| < 2^127
-    // k2 = c1*|B1| - c2*B2 in 256-bit two's complement
-    uint32_t k2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    uint32_t u6[6];
-    mul_wide<2, 4>(c1, GLV_B1ABS, u6);
-    for (int i = 0; i < 6; i++) k2[i] = u6[i];
-    mul_wide<4, 2>(c2, GLV_B2, u6);
-    uint32_t u6e[8] = {u6[0], u6[1], u6[2], u6[3], u6[4], u6[5], 0, 0};
-    sub_n<8>(k2, u6e);
-    neg2 = k2[7] >> 31;
-    if (neg2) neg_n<8>(k2);
-    for (int i = 0; i < 4; i++) m2[i] = k2[i];  // |k2| < 2^127
-}
-
-// ---- kernel 1: GLV split + signed window decomposition --------------------
-// val word: bits 0..29 point index, bit 30 = phi half (multiply base.x by
-// beta in bucket accumulation), bit 31 = negate point.
+// ---- kernel 1: signed window decomposition --------------------------------
 __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
                              uint32_t nbatch, int canonical,
                              uint32_t* __restrict__ keys,
@@ -145,41 +57,35 @@ __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
         fp256 s;
         ff_from_bytes(s, scalars + 32 * (b * n + i));
         if (!canonical) ff_from_mont<Fr>(s, s);
-        uint32_t mags[2][4], negs[2];
-        glv_decompose(s, mags[0], negs[0], mags[1], negs[1]);
-        for (uint32_t h = 0; h < MSM_HALVES; h++) {
-            uint32_t carry = 0;
-            for (int w = 0; w < MSM_NWIN; w++) {
-                const uint32_t bit0 = (uint32_t)w * MSM_WBITS;
-                const uint32_t li = bit0 >> 5, sh = bit0 & 31;
-                uint64_t pair = (uint64_t)mags[h][li];
-                if (li < 3) pair |= (uint64_t)mags[h][li + 1] << 32;
-                uint32_t d =
-                    ((uint32_t)(pair >> sh) & ((1u << MSM_WBITS) - 1)) + carry;
-                const uint32_t half = 1u << (MSM_WBITS - 1);
-                const uint32_t full = 1u << MSM_WBITS;
-                uint32_t key, val = (uint32_t)i | (h << 30);
-                uint32_t dneg = 0;
-                if (d == 0) {
-                    carry = 0;
-                    key = skip_key;
-                } else if (d <= half) {  // positive digit, magnitude d
-                    carry = 0;
-                    key = b * MSM_NB_TOTAL + w * MSM_BPW + (d - 1);
-                } else if (d == full) {  // max digit + carry out
-                    carry = 1;
-                    key = skip_key;
-                } else {  // negative digit, magnitude 2^c - d
-                    carry = 1;
-                    key = b * MSM_NB_TOTAL + w * MSM_BPW + (full - d - 1);
-                    dneg = 1;
-                }
-                if (dneg ^ negs[h]) val |= 0x80000000u;
-                const uint64_t slot =
-                    ((uint64_t)(b * MSM_HALVES + h) * MSM_NWIN + w) * n + i;
-                keys[slot] = key;
-                vals[slot] = val;
+        uint32_t carry = 0;
+        for (int w = 0; w < MSM_NWIN; w++) {
+            // c-bit window w of the canonical scalar (may cross limbs)
+            const uint32_t bit0 = (uint32_t)w * MSM_WBITS;
+            const uint32_t li = bit0 >> 5, sh = bit0 & 31;
+            uint64_t pair = (uint64_t)s.l[li];
+            if (li < 7) pair |= (uint64_t)s.l[li + 1] << 32;
+            uint32_t d =
+                ((uint32_t)(pair >> sh) & ((1u << MSM_WBITS) - 1)) + carry;
+            const uint32_t half = 1u << (MSM_WBITS - 1);
+            const uint32_t full = 1u << MSM_WBITS;
+            uint32_t key, val = (uint32_t)i;
+            if (d == 0) {
+                carry = 0;
+                key = skip_key;
+            } else if (d <= half) {  // positive digit, magnitude d
+                carry = 0;
+                key = b * MSM_NB_TOTAL + w * MSM_BPW + (d - 1);
+            } else if (d == full) {  // max digit + carry: digit 0, carry out
+                carry = 1;
+                key = skip_key;
+            } else {  // negative digit, magnitude 2^c - d
+                carry = 1;
+                key = b * MSM_NB_TOTAL + w * MSM_BPW + (full - d - 1);
+                val |= 0x80000000u;
             }
+            const uint64_t slot = ((uint64_t)b * MSM_NWIN + w) * n + i;
+            keys[slot] = key;
+            vals[slot] = val;
         }
     }
 }
@@ -240,15 +146,9 @@ PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
         }
         if (kj < nb_total) {
             uint32_t v = vals[j];
-            // plain (cached) loads: bases are re-read by all 16 entries'
-            // windows, so L1/L2 residency pays (non-temporal loads measured
-            // 5% slower)
-            g1_affine p = bases[v & 0x3fffffffu];
-            if (v & 0x40000000u) {  // phi half: (x, y) -> (beta*x, y)
-                fp256 beta;
-                for (int q = 0; q < 8; q++) beta.l[q] = GLV_BETA_MONT[q];
-                ff_mul<Fq>(p.x, p.x, beta);
-            }
+            // plain (cached) loads: bases are re-read by all 16 windows, so
+            // L1/L2 residency pays (non-temporal loads measured 5% slower)
+            g1_affine p = bases[v & 0x7fffffffu];
             if (v & 0x80000000u) ff_neg<Fq>(p.y, p.y);
             g1j_madd_ip(acc, p);
         }
@@ -361,7 +261,7 @@ __global__ __launch_bounds__(WSUM_THREADS, 1) void k_window_sum(
 
 // ---- host orchestration ---------------------------------------------------
 static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
-    const uint64_t ent = (uint64_t)nbatch * MSM_ENT_PER * n;
+    const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
     const uint64_t nbt = (uint64_t)nbatch * MSM_NB_TOTAL;
     if (ds.ent_cap < ent) {
         for (void* p : {(void*)ds.d_keys_in, (void*)ds.d_keys_out,
@@ -427,7 +327,7 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
     }
     int rc = ensure_msm_scratch(ds, n, nbatch);
     if (rc) return rc;
-    const uint64_t ent = (uint64_t)nbatch * MSM_ENT_PER * n;
+    const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
     const uint32_t nbt = nbatch * MSM_NB_TOTAL;
     const int canonical = (flags & SPECTRE_SCALARS_CANONICAL) ? 1 : 0;
     hipStream_t st = ds.stream;

@@ -13,7 +13,7 @@ _LIB = os.path.join(_HERE, "libspectre_gpu.so")
 SCALARS_MONTGOMERY = 0
 SCALARS_CANONICAL = 1
 WINDOW_BITS = 16  # = SPECTRE_MSM_WINDOW_BITS
-NUM_WINDOWS = 8   # = SPECTRE_MSM_NUM_WINDOWS (GLV halves)
+NUM_WINDOWS = 16  # = SPECTRE_MSM_NUM_WINDOWS
 NUM_BUCKETS = NUM_WINDOWS * (1 << (WINDOW_BITS - 1))
 PARTIALS_BYTES = NUM_WINDOWS * 96  # per-shard Jacobian window sums
 
