@@ -195,61 +195,42 @@ PT_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
     buckets[b] = acc;
 }
 
-// ---- kernel 5: wave-cooperative weighted chunk scan -----------------------
-// Window-local bucket j has multiplier (j+1); for the 64-bucket chunk at
-// window offset m this kernel computes
-//   P = sum_j (j+1) B[m+j] + m * sum_j B[m+j]
-// with one WAVE per chunk: an inclusive shuffle suffix-scan (6 EC-add
-// levels; sum over lanes of the suffixes = the weighted sum), a shuffle
-// reduction, and one 15-bit double-and-add on lane 0. The earlier
-// serial-per-thread version ran at <=1 wave/SIMD and was a pure latency
-// floor (~0.7 ms); wave cooperation puts 8 waves/SIMD of log-depth work on
-// the chip instead.
-#define SCAN_CHUNK 64u
-__device__ __forceinline__ void shfl_down_jac(g1_jac& o, const g1_jac& v,
-                                              int delta) {
-    const int* s = (const int*)&v;
-    int* d = (int*)&o;
-    for (int q = 0; q < 24; q++) d[q] = __shfl_down(s[q], delta, 64);
-}
-
-PT_KERNEL void k_window_scan(const g1_jac* __restrict__ buckets,
-                             uint32_t total_chunks,
-                             g1_jac* __restrict__ out) {
-    const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
-    const uint32_t wave = gid / 64, lane = gid & 63;
-    if (wave >= total_chunks) return;  // wave-uniform exit
-    const uint32_t chunks_pw = MSM_BPW / SCAN_CHUNK;
-    __shared__ g1_jac Tpark[THREADS / 64];
-    g1_jac acc = buckets[(uint64_t)wave * SCAN_CHUNK + lane];
-    // inclusive suffix scan: acc_l = sum_{j>=l} B_j
-    for (int d = 1; d < 64; d <<= 1) {
-        g1_jac other;
-        shfl_down_jac(other, acc, d);
-        if (lane + d < 64) g1j_add_ip(acc, other);
+// ---- kernel 5: weighted chunk reduction -----------------------------------
+// Window-local bucket j has multiplier (j+1). For the chunk covering
+// window-local buckets [m, m+CHUNK):
+//   sum_j (j+1) B[m+j]  (suffix running sums)  +  m * sum_j B[m+j]
+// (double-and-add; m < 2^(WBITS-1)). Uniform in the flattened
+// (batch*NWIN + w) window index, so batching needs no changes here.
+PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
+                               uint32_t total_chunks,
+                               g1_jac* __restrict__ out) {
+    const uint32_t nchunks_pw = MSM_BPW / MSM_CHUNK;
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= total_chunks) return;
+    uint32_t cw = t % nchunks_pw;
+    const g1_jac* B = buckets + (uint64_t)t * MSM_CHUNK;
+    // Phase order keeps at most TWO Jacobian accumulators live (3 would push
+    // past 256 VGPRs and spill): accW is parked in `out` before the
+    // double-and-add, then re-loaded for the final add.
+    g1_jac accT, accW;
+    g1j_set_inf(accT);
+    g1j_set_inf(accW);
+    for (int j = MSM_CHUNK - 1; j >= 0; j--) {
+        g1j_add_ip(accT, B[j]);
+        g1j_add_ip(accW, accT);
     }
-    // lane 0 holds the chunk total T after the scan; park it in LDS so it
-    // is not live (24 VGPRs) across the 6-level reduction below
-    if (lane == 0) Tpark[threadIdx.x / 64] = acc;
-    // weighted sum W = sum_l acc_l (reduce to lane 0)
-    for (int d = 32; d >= 1; d >>= 1) {
-        g1_jac other;
-        shfl_down_jac(other, acc, d);
-        if (lane + d < 64) g1j_add_ip(acc, other);
-    }
-    if (lane == 0) {
-        const uint32_t m = (wave % chunks_pw) * SCAN_CHUNK;
-        g1_jac T = Tpark[threadIdx.x / 64];
-        if (m && !g1j_is_inf(T)) {
-            g1_jac sc;
-            g1j_set_inf(sc);
-            for (int bit = MSM_WBITS - 1; bit >= 0; bit--) {
-                g1j_dbl_ip(sc);
-                if ((m >> bit) & 1) g1j_add_ip(sc, T);
-            }
-            g1j_add_ip(acc, sc);
+    out[t] = accW;  // park; accW dead
+    const uint32_t m = cw * MSM_CHUNK;
+    if (m && !g1j_is_inf(accT)) {
+        g1_jac acc;
+        g1j_set_inf(acc);
+        for (int bit = MSM_WBITS - 1; bit >= 0; bit--) {
+            g1j_dbl_ip(acc);
+            if ((m >> bit) & 1) g1j_add_ip(acc, accT);
         }
-        out[wave] = acc;
+        g1_jac w = out[t];  // accT dead; (w, acc) live
+        g1j_add_ip(w, acc);
+        out[t] = w;
     }
 }
 
@@ -258,11 +239,11 @@ PT_KERNEL void k_window_scan(const g1_jac* __restrict__ buckets,
 // threads grid-stride their share, then a pairwise LDS tree — serial depth
 // ~16 adds in ONE launch (a 3-level cascade of tiny grids was ~0.8 ms of
 // launch+latency overhead).
-#define WSUM_THREADS 256
-__global__ __launch_bounds__(WSUM_THREADS, 2) void k_window_sum(
+#define WSUM_THREADS 1024
+__global__ __launch_bounds__(WSUM_THREADS, 1) void k_window_sum(
     const g1_jac* __restrict__ in, g1_jac* __restrict__ out) {
     __shared__ g1_jac lds[WSUM_THREADS / 2];
-    const uint32_t per_win = MSM_BPW / SCAN_CHUNK;
+    const uint32_t per_win = MSM_BPW / MSM_CHUNK;
     const uint32_t w = blockIdx.x;
     const uint32_t t = threadIdx.x;
     g1_jac acc;
@@ -306,7 +287,7 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
         HIP_TRY(hipMalloc(&ds.d_offsets, (nbt + 1) * 4));
         HIP_TRY(hipMalloc(&ds.d_buckets, nbt * sizeof(g1_jac)));
         HIP_TRY(hipMalloc(&ds.d_red,
-                          (nbt / SCAN_CHUNK + nbatch * MSM_NWIN) * sizeof(g1_jac)));
+                          (nbt / MSM_CHUNK + nbatch * MSM_NWIN) * sizeof(g1_jac)));
         ds.nb_cap = nbt;
     }
     size_t sort_need = 0;
@@ -383,12 +364,12 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                        dim3(THREADS), 0, st, ds.d_offsets, ds.d_firstK,
                        ds.d_firstP, ds.d_lastK, ds.d_lastP, nbt, ds.d_buckets);
     STAMP(4);
-    const uint32_t nchunks = nbt / SCAN_CHUNK;
+    const uint32_t nchunks = nbt / MSM_CHUNK;
     g1_jac* red0 = ds.d_red;
     g1_jac* red1 = ds.d_red + nchunks;
-    hipLaunchKernelGGL(k_window_scan,
-                       dim3((nchunks * 64 + THREADS - 1) / THREADS),
-                       dim3(THREADS), 0, st, ds.d_buckets, nchunks, red0);
+    hipLaunchKernelGGL(k_window_chunks,
+                       dim3((nchunks + THREADS - 1) / THREADS), dim3(THREADS),
+                       0, st, ds.d_buckets, nchunks, red0);
     STAMP(5);
     hipLaunchKernelGGL(k_window_sum, dim3(nbatch * MSM_NWIN),
                        dim3(WSUM_THREADS), 0, st, red0, red1);
