@@ -239,8 +239,8 @@ PT_KERNEL void k_window_chunks(const g1_jac* __restrict__ buckets,
 // threads grid-stride their share, then a pairwise LDS tree — serial depth
 // ~16 adds in ONE launch (a 3-level cascade of tiny grids was ~0.8 ms of
 // launch+latency overhead).
-#define WSUM_THREADS 1024
-__global__ __launch_bounds__(WSUM_THREADS, 1) void k_window_sum(
+#define WSUM_THREADS 256
+__global__ __launch_bounds__(WSUM_THREADS, 2) void k_window_sum(
     const g1_jac* __restrict__ in, g1_jac* __restrict__ out) {
     __shared__ g1_jac lds[WSUM_THREADS / 2];
     const uint32_t per_win = MSM_BPW / MSM_CHUNK;
