@@ -56,19 +56,19 @@ __global__ __launch_bounds__(256, 2) void k_mul_chain(const g1_affine* pts,
 static double time_kernel(void (*fn)(const g1_affine*, g1_jac*, int),
                           const g1_affine* pts, g1_jac* out, int nwork,
                           int blocks) {
-    hipEvent_t e0, e1;
-    hipEventCreate(&e0);
-    hipEventCreate(&e1);
+    (void)hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
     hipLaunchKernelGGL(fn, dim3(blocks), dim3(256), 0, 0, pts, out, nwork);
-    hipDeviceSynchronize();
-    hipEventRecord(e0);
+    (void)hipDeviceSynchronize();
+    (void)hipEventRecord(e0);
     hipLaunchKernelGGL(fn, dim3(blocks), dim3(256), 0, 0, pts, out, nwork);
-    hipEventRecord(e1);
-    hipEventSynchronize(e1);
+    (void)hipEventRecord(e1);
+    (void)hipEventSynchronize(e1);
     float ms;
-    hipEventElapsedTime(&ms, e0, e1);
-    hipEventDestroy(e0);
-    hipEventDestroy(e1);
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
     return ms;
 }
 
@@ -77,9 +77,9 @@ int main() {
     // throughput probe; formulas don't branch on curve membership)
     g1_affine* pts;
     g1_jac* out;
-    hipMalloc(&pts, 1024 * sizeof(g1_affine));
-    hipMalloc(&out, (size_t)4 * 1024 * 1024 * sizeof(g1_jac));
-    hipMemset(pts, 0x5a, 1024 * sizeof(g1_affine));
+    (void)hipMalloc(&pts, 1024 * sizeof(g1_affine));
+    (void)hipMalloc(&out, (size_t)4 * 1024 * 1024 * sizeof(g1_jac));
+    (void)hipMemset(pts, 0x5a, 1024 * sizeof(g1_affine));
     for (int nwork : {256 * 1024, 1024 * 1024}) {
         int blocks = nwork / 256;
         double ms1 = time_kernel(k_madd_chain<1>, pts, out, nwork, blocks);
@@ -97,20 +97,20 @@ int main() {
            (double)nwork * ITERS * 16 / msm / 1e3);
     // raw mad64 rate
     {
-        hipEvent_t e0, e1;
-        hipEventCreate(&e0);
-        hipEventCreate(&e1);
+        (void)hipEvent_t e0, e1;
+        (void)hipEventCreate(&e0);
+        (void)hipEventCreate(&e1);
         uint32_t iters = 4096;
         hipLaunchKernelGGL(k_mad64_rate, dim3(nwork / 256), dim3(256), 0, 0,
                            (uint64_t*)out, iters);
-        hipDeviceSynchronize();
-        hipEventRecord(e0);
+        (void)hipDeviceSynchronize();
+        (void)hipEventRecord(e0);
         hipLaunchKernelGGL(k_mad64_rate, dim3(nwork / 256), dim3(256), 0, 0,
                            (uint64_t*)out, iters);
-        hipEventRecord(e1);
-        hipEventSynchronize(e1);
+        (void)hipEventRecord(e1);
+        (void)hipEventSynchronize(e1);
         float ms;
-        hipEventElapsedTime(&ms, e0, e1);
+        (void)hipEventElapsedTime(&ms, e0, e1);
         double mads = (double)nwork * iters * 4;
         printf("v_mad_u64_u32 (4 indep chains): %.3f ms  %.2f T mad64/s "
                "(peak-if-2cyc = 78.6T, 4cyc = 39.3T, 8cyc = 19.7T)\n", ms,
