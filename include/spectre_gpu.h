@@ -124,6 +124,23 @@ int spectre_gpu_ntt_fr_device(spectre_gpu_ctx*, int dev, void* d_data,
                               uint32_t log_n, const uint8_t omega[32],
                               int inverse, const uint8_t* coset_gen);
 
+/* ---- pointwise Fr vector ops (device buffers, Montgomery form) ----------
+ * The quotient phase evaluates gate expressions pointwise between the
+ * iFFT and coset-FFT passes; these ops let a patched evaluator keep
+ * polynomials device-resident. out may alias a or b.
+ *   op 0: out = a + b        op 1: out = a - b       op 2: out = a * b
+ *   op 3: out = a * c        op 4: out = a + c * b   (c = one Fr element)
+ */
+#define SPECTRE_VEC_ADD 0
+#define SPECTRE_VEC_SUB 1
+#define SPECTRE_VEC_MUL 2
+#define SPECTRE_VEC_SCALE 3
+#define SPECTRE_VEC_ADD_SCALED 4
+int spectre_gpu_fr_vec_op(spectre_gpu_ctx*, int dev, int op,
+                          const void* d_a, const void* d_b /* NULL for op 3 */,
+                          const uint8_t c[32] /* NULL for ops 0-2 */,
+                          void* d_out, uint64_t n);
+
 /* ---- device memory helpers (for C callers and the bench harness) ------- */
 int spectre_gpu_malloc(spectre_gpu_ctx*, int dev, size_t bytes, void** d_ptr);
 int spectre_gpu_free(spectre_gpu_ctx*, int dev, void* d_ptr);

@@ -454,6 +454,23 @@ int spectre_gpu_ntt_fr(spectre_gpu_ctx* ctx, uint8_t* data, uint32_t log_n,
     return rc;
 }
 
+int spectre_gpu_fr_vec_op(spectre_gpu_ctx* ctx, int dev, int op,
+                          const void* d_a, const void* d_b, const uint8_t* c,
+                          void* d_out, uint64_t n) {
+    if (check_dev(ctx, dev)) return -1;
+    if (op < 0 || op > 4 || !d_a || !d_out ||
+        (op <= 2 && !d_b) || (op == 4 && !d_b) || (op >= 3 && !c)) {
+        set_err("fr_vec_op: bad op/arguments");
+        return -1;
+    }
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    fp256 cv;
+    if (c) ff_from_bytes(cv, c);
+    return fr_vec_op_device(ctx, dev, op, (const fp256*)d_a,
+                            (const fp256*)d_b, c ? &cv : nullptr,
+                            (fp256*)d_out, n);
+}
+
 // ---------------------------------------------------------------- memory
 int spectre_gpu_malloc(spectre_gpu_ctx* ctx, int dev, size_t bytes,
                        void** d_ptr) {

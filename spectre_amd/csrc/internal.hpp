@@ -112,6 +112,10 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            g1_jac* winsums_host, double* stage_ms = nullptr,
                            bool sync = true);
 
+// ntt.hip — pointwise Fr vector op on device buffers (synchronizes).
+int fr_vec_op_device(spectre_gpu_ctx* ctx, int dev, int op, const fp256* d_a,
+                     const fp256* d_b, const fp256* c, fp256* d_out,
+                     uint64_t n);
 // ntt.hip — in-place NTT on a device buffer (synchronizes the stream).
 int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
                const fp256& omega, int inverse, const fp256* coset_gen);

@@ -241,6 +241,44 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_axis(
     }
 }
 
+// ---- pointwise Fr vector ops (quotient-phase gate-eval glue) --------------
+__global__ void k_fr_vec_op(int op, const fp256* __restrict__ a,
+                            const fp256* __restrict__ b, fp256 c,
+                            fp256* __restrict__ out, uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    fp256 r, va = a[i];
+    switch (op) {
+        case SPECTRE_VEC_ADD: ff_add<Fr>(r, va, b[i]); break;
+        case SPECTRE_VEC_SUB: ff_sub<Fr>(r, va, b[i]); break;
+        case SPECTRE_VEC_MUL: ff_mul<Fr>(r, va, b[i]); break;
+        case SPECTRE_VEC_SCALE: ff_mul<Fr>(r, va, c); break;
+        default: {  // ADD_SCALED
+            fp256 t;
+            ff_mul<Fr>(t, b[i], c);
+            ff_add<Fr>(r, va, t);
+        }
+    }
+    out[i] = r;
+}
+
+int fr_vec_op_device(spectre_gpu_ctx* ctx, int dev, int op, const fp256* d_a,
+                     const fp256* d_b, const fp256* c, fp256* d_out,
+                     uint64_t n) {
+    DeviceState& ds = ctx->devs[dev];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    fp256 cv;
+    ff_set_zero(cv);
+    if (c) cv = *c;
+    hipLaunchKernelGGL(k_fr_vec_op,
+                       dim3((uint32_t)((n + THREADS - 1) / THREADS)),
+                       dim3(THREADS), 0, ds.stream, op, d_a, d_b, cv, d_out,
+                       n);
+    HIP_TRY(hipStreamSynchronize(ds.stream));
+    HIP_TRY(hipGetLastError());
+    return 0;
+}
+
 // ---------------------------------------------------------------- host side
 static void host_pow_u32(fp256& o, const fp256& a, uint32_t e) {
     ff_pow_u32<Fr>(o, a, e);

@@ -84,6 +84,11 @@ def load_library() -> ctypes.CDLL:
         c.c_void_p, c.c_int, c.c_void_p, c.c_uint32, c.c_void_p, c.c_int,
         c.c_void_p,
     ]
+    lib.spectre_gpu_fr_vec_op.restype = c.c_int
+    lib.spectre_gpu_fr_vec_op.argtypes = [
+        c.c_void_p, c.c_int, c.c_int, c.c_void_p, c.c_void_p, c.c_void_p,
+        c.c_void_p, c.c_uint64,
+    ]
     lib.spectre_gpu_malloc.restype = c.c_int
     lib.spectre_gpu_malloc.argtypes = [c.c_void_p, c.c_int, c.c_size_t,
                                        c.POINTER(c.c_void_p)]
@@ -251,6 +256,18 @@ class SpectreGpu:
             self._ctx, dev, ctypes.c_void_p(d_data), log_n, om,
             1 if inverse else 0, cg)
         self._check(rc, "ntt_fr_device")
+
+    VEC_ADD, VEC_SUB, VEC_MUL, VEC_SCALE, VEC_ADD_SCALED = range(5)
+
+    def fr_vec_op(self, op: int, d_a: int, d_b: int | None, c: bytes | None,
+                  d_out: int, n: int, dev: int = 0) -> None:
+        """Pointwise Fr vector op on device buffers (Montgomery form)."""
+        cc = (ctypes.c_uint8 * 32).from_buffer_copy(c) if c else None
+        rc = self._lib.spectre_gpu_fr_vec_op(
+            self._ctx, dev, op, ctypes.c_void_p(d_a),
+            ctypes.c_void_p(d_b) if d_b else None, cc,
+            ctypes.c_void_p(d_out), n)
+        self._check(rc, "fr_vec_op")
 
     # ---- device memory helpers ----
     def malloc(self, nbytes: int, dev: int = 0) -> int:

@@ -15,7 +15,7 @@ SYMBOLS = [
     "spectre_gpu_msm_g1_device", "spectre_gpu_msm_g1_shard_device",
     "spectre_gpu_msm_g1_combine", "spectre_gpu_msm_g1_shard_device_timed",
     "spectre_gpu_msm_g1_batch", "spectre_gpu_msm_g1_batch_device",
-    "spectre_gpu_ntt_fr",
+    "spectre_gpu_ntt_fr", "spectre_gpu_fr_vec_op",
     "spectre_gpu_ntt_fr_device", "spectre_gpu_malloc", "spectre_gpu_free",
     "spectre_gpu_upload", "spectre_gpu_download", "spectre_gpu_synchronize",
 ]
