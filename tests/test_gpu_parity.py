@@ -410,3 +410,36 @@ def test_fr_vec_ops_vs_oracle(gpu, oracle):
     assert got == want
     for p in (d_a, d_b, d_o):
         gpu.free(p)
+
+
+def test_msm_randomized_sizes(gpu, oracle):
+    """Seeded sweep over awkward sizes (odd, prime-ish, off-by-one around
+    thread/partition boundaries) x scalar encodings — edge insurance for the
+    equal-work partitioning and boundary-fixup logic."""
+    import random
+    rng = random.Random(20260915)
+    sizes = [1, 2, 63, 64, 65, 255, 1000003 % 4096, 4095, 4097,
+             rng.randrange(1, 20000), rng.randrange(1, 20000)]
+    for idx, n in enumerate(sizes):
+        sc, bs = oracle.gen_msm_inputs(n, 7000 + idx, fast=True)
+        want = oracle.msm(bs, sc, n)
+        assert gpu.msm(bs, sc, n) == want, (n, "canonical")
+        scm = b"".join(
+            oracle.fr_from_canonical(sc[32 * i:32 * i + 32])
+            for i in range(n))
+        assert gpu.msm(bs, scm, n, canonical=False) == want, (n, "mont")
+
+
+def test_msm_batch_randomized(gpu, oracle):
+    import random
+    rng = random.Random(99)
+    for trial in range(3):
+        n = rng.randrange(100, 5000)
+        nb = rng.randrange(2, 7)
+        _, bs = oracle.gen_msm_inputs(n, 7100 + trial, fast=True)
+        scal, singles = b"", []
+        for b in range(nb):
+            sc, _ = oracle.gen_msm_inputs(n, 7200 + 10 * trial + b, fast=True)
+            scal += sc
+            singles.append(oracle.msm(bs, sc, n))
+        assert gpu.msm_batch(bs, scal, nb, n) == singles, (n, nb)
