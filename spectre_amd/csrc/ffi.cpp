@@ -66,7 +66,8 @@ void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
               (void*)ds.d_buckets, (void*)ds.d_firstK, (void*)ds.d_lastK,
               (void*)ds.d_firstP, (void*)ds.d_lastP,
               (void*)ds.d_red, (void*)ds.d_scalars,
-              (void*)ds.d_bases, (void*)ds.d_ntt_tmp, (void*)ds.d_cosetA})
+              (void*)ds.d_bases, (void*)ds.d_ntt_tmp, (void*)ds.d_ntt_io,
+              (void*)ds.d_cosetA})
             if (p) (void)hipFree(p);
         for (int i = 0; i < 2; i++) {
             if (ds.h_stage[i]) (void)hipHostFree(ds.h_stage[i]);
@@ -295,7 +296,7 @@ int spectre_gpu_msm_g1_batch(spectre_gpu_ctx* ctx, uint64_t bases_id,
     }
     g1_affine* d_b = nullptr;
     if (bases_id != 0) {
-        auto key = std::make_pair(bases_id, n * 16 + 1);
+        std::array<uint64_t, 3> key{bases_id, n, 1};
         auto it = ds.bases_cache.find(key);
         if (it != ds.bases_cache.end()) {
             d_b = it->second.d_ptr;
@@ -331,6 +332,69 @@ int spectre_gpu_msm_g1_batch(spectre_gpu_ctx* ctx, uint64_t bases_id,
                                            n, flags, out_affine);
 }
 
+// upload + enqueue one device's shard of an n-point MSM split num_gpus ways
+// (returns with async work in flight on that device's stream).
+static int msm_multi_enqueue_one(spectre_gpu_ctx* ctx, int d, int num_gpus,
+                                 uint64_t bases_id, const uint8_t* bases,
+                                 const uint8_t* scalars, uint64_t n,
+                                 uint32_t flags, g1_jac* wins_out) {
+    DeviceState& ds = ctx->devs[d];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    const uint64_t lo = n * d / num_gpus, hi = n * (d + 1) / num_gpus;
+    const uint64_t m = hi - lo;
+    // scalars: plain scratch upload
+    if (ds.scal_cap < m * 32) {
+        if (ds.d_scalars) { (void)hipFree(ds.d_scalars); ds.d_scalars = nullptr; }
+        ds.scal_cap = 0;
+        HIP_TRY(hipMalloc(&ds.d_scalars, m * 32));
+        ds.scal_cap = m * 32;
+    }
+    {
+        int rc = staged_upload(ds, ds.d_scalars, scalars + lo * 32, m * 32);
+        if (rc) return rc;
+    }
+    // bases: cached per (bases_id, n, num_gpus) — this device's chunk layout
+    // depends on all three, so all three form the cache key.
+    g1_affine* d_b = nullptr;
+    if (bases_id != 0) {
+        std::array<uint64_t, 3> key{bases_id, n, (uint64_t)num_gpus};
+        auto it = ds.bases_cache.find(key);
+        if (it != ds.bases_cache.end()) {
+            d_b = it->second.d_ptr;
+        } else {
+            if (!bases) {
+                set_err("bases_id %llu not cached and bases is NULL",
+                        (unsigned long long)bases_id);
+                return -4;
+            }
+            CachedBases cb;
+            cb.n = m;
+            HIP_TRY(hipMalloc(&cb.d_ptr, m * sizeof(g1_affine)));
+            int rc = staged_upload(ds, cb.d_ptr, bases + lo * 64, m * 64);
+            if (rc) return rc;
+            ds.bases_cache.emplace(key, cb);
+            d_b = cb.d_ptr;
+        }
+    } else {
+        if (!bases) {
+            set_err("bases is NULL");
+            return -1;
+        }
+        if (ds.base_cap < m) {
+            if (ds.d_bases) { (void)hipFree(ds.d_bases); ds.d_bases = nullptr; }
+            ds.base_cap = 0;
+            HIP_TRY(hipMalloc(&ds.d_bases, m * sizeof(g1_affine)));
+            ds.base_cap = m;
+        }
+        int rc = staged_upload(ds, ds.d_bases, bases + lo * 64, m * 64);
+        if (rc) return rc;
+        d_b = ds.d_bases;
+    }
+    // enqueue without synchronizing so all shards run concurrently
+    return msm_batch_shard_device(ctx, d, d_b, ds.d_scalars, 1, m, flags,
+                                  wins_out, nullptr, /*sync=*/false);
+}
+
 int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
                        const uint8_t* bases, const uint8_t* scalars, uint64_t n,
                        uint32_t flags, int num_gpus, uint8_t out_affine[64]) {
@@ -354,61 +418,18 @@ int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
     }
     std::vector<g1_jac> wins((size_t)num_gpus * MSM_NWIN);
     for (int d = 0; d < num_gpus; d++) {
-        DeviceState& ds = ctx->devs[d];
-        HIP_TRY(hipSetDevice(ds.device_id));
-        const uint64_t lo = n * d / num_gpus, hi = n * (d + 1) / num_gpus;
-        const uint64_t m = hi - lo;
-        // scalars: plain scratch upload
-        if (ds.scal_cap < m * 32) {
-            if (ds.d_scalars) (void)hipFree(ds.d_scalars);
-            HIP_TRY(hipMalloc(&ds.d_scalars, m * 32));
-            ds.scal_cap = m * 32;
-        }
-        {
-            int rc = staged_upload(ds, ds.d_scalars, scalars + lo * 32, m * 32);
-            if (rc) return rc;
-        }
-        // bases: cached per (bases_id, n, num_gpus) — shard layout depends on
-        // all three, so they are all folded into the cache key.
-        g1_affine* d_b = nullptr;
-        if (bases_id != 0) {
-            auto key = std::make_pair(bases_id, n * 16 + (uint64_t)num_gpus);
-            auto it = ds.bases_cache.find(key);
-            if (it != ds.bases_cache.end()) {
-                d_b = it->second.d_ptr;
-            } else {
-                if (!bases) {
-                    set_err("bases_id %llu not cached and bases is NULL",
-                            (unsigned long long)bases_id);
-                    return -4;
-                }
-                CachedBases cb;
-                cb.n = m;
-                HIP_TRY(hipMalloc(&cb.d_ptr, m * sizeof(g1_affine)));
-                int rc = staged_upload(ds, cb.d_ptr, bases + lo * 64, m * 64);
-                if (rc) return rc;
-                ds.bases_cache.emplace(key, cb);
-                d_b = cb.d_ptr;
+        int rc = msm_multi_enqueue_one(ctx, d, num_gpus, bases_id, bases,
+                                       scalars, n, flags,
+                                       &wins[(size_t)d * MSM_NWIN]);
+        if (rc) {
+            // Devices <= d may have kernels and an async D2H into `wins`
+            // in flight; drain them before the vector goes out of scope.
+            for (int e = 0; e <= d; e++) {
+                (void)hipSetDevice(ctx->devs[e].device_id);
+                (void)hipStreamSynchronize(ctx->devs[e].stream);
             }
-        } else {
-            if (!bases) {
-                set_err("bases is NULL");
-                return -1;
-            }
-            if (ds.base_cap < m) {
-                if (ds.d_bases) (void)hipFree(ds.d_bases);
-                HIP_TRY(hipMalloc(&ds.d_bases, m * sizeof(g1_affine)));
-                ds.base_cap = m;
-            }
-            int rc = staged_upload(ds, ds.d_bases, bases + lo * 64, m * 64);
-            if (rc) return rc;
-            d_b = ds.d_bases;
+            return rc;
         }
-        // enqueue without synchronizing so all shards run concurrently
-        int rc = msm_batch_shard_device(ctx, d, d_b, ds.d_scalars, 1, m,
-                                        flags, &wins[(size_t)d * MSM_NWIN],
-                                        nullptr, /*sync=*/false);
-        if (rc) return rc;
     }
     for (int d = 0; d < num_gpus; d++) {
         HIP_TRY(hipSetDevice(ctx->devs[d].device_id));
@@ -443,14 +464,19 @@ int spectre_gpu_ntt_fr(spectre_gpu_ctx* ctx, uint8_t* data, uint32_t log_n,
     DeviceState& ds = ctx->devs[0];
     HIP_TRY(hipSetDevice(ds.device_id));
     const uint64_t n = 1ull << log_n;
-    void* d_buf = nullptr;
-    HIP_TRY(hipMalloc(&d_buf, n * 32));
-    int rc = staged_upload(ds, d_buf, data, n * 32);
+    // cached IO buffer: the Rust seam calls this entry ~40-60x per proof and
+    // must not pay a (up to 1 GB) hipMalloc/hipFree per transform.
+    if (ds.ntt_io_cap < n) {
+        if (ds.d_ntt_io) { (void)hipFree(ds.d_ntt_io); ds.d_ntt_io = nullptr; }
+        ds.ntt_io_cap = 0;
+        HIP_TRY(hipMalloc(&ds.d_ntt_io, n * 32));
+        ds.ntt_io_cap = n;
+    }
+    int rc = staged_upload(ds, ds.d_ntt_io, data, n * 32);
     if (rc == 0)
-        rc = spectre_gpu_ntt_fr_device(ctx, 0, d_buf, log_n, omega, inverse,
-                                       coset_gen);
-    if (rc == 0) rc = staged_download(ds, data, d_buf, n * 32);
-    (void)hipFree(d_buf);
+        rc = spectre_gpu_ntt_fr_device(ctx, 0, ds.d_ntt_io, log_n, omega,
+                                       inverse, coset_gen);
+    if (rc == 0) rc = staged_download(ds, data, ds.d_ntt_io, n * 32);
     return rc;
 }
 

@@ -63,13 +63,19 @@ struct DeviceState {
     size_t scal_cap = 0;  // bytes
     g1_affine* d_bases = nullptr;
     size_t base_cap = 0;  // points
-    std::map<std::pair<uint64_t, uint64_t>, CachedBases> bases_cache;  // (id, n)
+    // key = (bases_id, n, shard layout): layout is num_gpus for the sharded
+    // path (this device's contiguous chunk of an n-point set split num_gpus
+    // ways) and 1 for the full set (batch path == num_gpus=1 shard).
+    std::map<std::array<uint64_t, 3>, CachedBases> bases_cache;
     // ---- pinned staging for host-pointer uploads/downloads ----
     uint8_t* h_stage[2] = {nullptr, nullptr};
     hipEvent_t stage_ev[2] = {nullptr, nullptr};
     // ---- NTT scratch ----
     fp256* d_ntt_tmp = nullptr;
     size_t ntt_cap = 0;  // elements
+    fp256* d_ntt_io = nullptr;  // cached device buffer for host-pointer NTTs
+    size_t ntt_io_cap = 0;      // elements (the Rust seam calls this path
+                                // ~40-60x per proof; no per-call hipMalloc)
     fp256* d_cosetA = nullptr;  // coset base tables (n2 resp. n1 entries)
     fp256* d_cosetB = nullptr;
     size_t coset_cap = 0;

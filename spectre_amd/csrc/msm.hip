@@ -264,11 +264,15 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
     const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
     const uint64_t nbt = (uint64_t)nbatch * MSM_NB_TOTAL;
     if (ds.ent_cap < ent) {
-        for (void* p : {(void*)ds.d_keys_in, (void*)ds.d_keys_out,
-                        (void*)ds.d_vals_in, (void*)ds.d_vals_out,
-                        (void*)ds.d_firstK, (void*)ds.d_lastK,
-                        (void*)ds.d_firstP, (void*)ds.d_lastP})
-            if (p) (void)hipFree(p);
+        // free-and-null before reallocating: a failed hipMalloc below must
+        // not leave dangling pointers with a stale capacity (the caps are
+        // only advanced after every allocation in the group succeeds).
+        ds.ent_cap = 0;
+        for (void** p : {(void**)&ds.d_keys_in, (void**)&ds.d_keys_out,
+                         (void**)&ds.d_vals_in, (void**)&ds.d_vals_out,
+                         (void**)&ds.d_firstK, (void**)&ds.d_lastK,
+                         (void**)&ds.d_firstP, (void**)&ds.d_lastP})
+            if (*p) { (void)hipFree(*p); *p = nullptr; }
         HIP_TRY(hipMalloc(&ds.d_keys_in, ent * 4));
         HIP_TRY(hipMalloc(&ds.d_keys_out, ent * 4));
         HIP_TRY(hipMalloc(&ds.d_vals_in, ent * 4));
@@ -281,9 +285,10 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
         ds.ent_cap = ent;
     }
     if (ds.nb_cap < nbt) {
-        for (void* p : {(void*)ds.d_offsets, (void*)ds.d_buckets,
-                        (void*)ds.d_red})
-            if (p) (void)hipFree(p);
+        ds.nb_cap = 0;
+        for (void** p : {(void**)&ds.d_offsets, (void**)&ds.d_buckets,
+                         (void**)&ds.d_red})
+            if (*p) { (void)hipFree(*p); *p = nullptr; }
         HIP_TRY(hipMalloc(&ds.d_offsets, (nbt + 1) * 4));
         HIP_TRY(hipMalloc(&ds.d_buckets, nbt * sizeof(g1_jac)));
         HIP_TRY(hipMalloc(&ds.d_red,
@@ -318,6 +323,16 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
     if (nbatch == 0 || nbatch > SPECTRE_MSM_MAX_BATCH) {
         set_err("msm: nbatch %u out of range [1,%d]", nbatch,
                 SPECTRE_MSM_MAX_BATCH);
+        return -3;
+    }
+    // Sort entries and segment offsets are indexed with uint32: at exactly
+    // nbatch*NWIN*n == 2^32 the final offset off[nb_total] wraps to 0 and the
+    // result is silently wrong. Reject; callers split such jobs (e.g. batch
+    // 32 x 2^23 -> 2 x 16).
+    if ((uint64_t)nbatch * MSM_NWIN * n >= (1ull << 32)) {
+        set_err("msm: nbatch*%d*n = %llu exceeds 2^32 sort-entry limit "
+                "(split the batch or shard the MSM)",
+                MSM_NWIN, (unsigned long long)((uint64_t)nbatch * MSM_NWIN * n));
         return -3;
     }
     if (n == 0) {
