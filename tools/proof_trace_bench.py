@@ -5,11 +5,14 @@ single sync-step proof issues through the best_multiexp/best_fft seam,
 against the same sequence on the CPU oracle (the ">=10x vs host CPU"
 check at proof granularity, not just per-kernel).
 
-Call counts are the structural estimates of SURVEY.md §3a for the pinned
-k=20 config (15 advice + 2 lookup-advice columns, lookup_bits 19, extended
-domain 2^22); exact counts await the instrumented Rust shim
-in a cargo-capable environment, which logs them for free. Witness data is
-synthetic (the arithmetic is shape-dependent only).
+Call counts are DERIVED from the PSE create_proof source structure for the
+pinned configs — full derivation with per-stage formulas in CALLCOUNTS.md
+(sync-step k=20: 19 advice columns incl. the SpreadConfig pair, 3 lookup
+arguments, 11 permutation z-chunks at degree 4 => 45 MSM(2^20) + 40
+iFFT(2^20) + 40 coset-FFT(2^22) + 1 extended icoset). committee24 remains
+an estimate (zkevm-hashes SHA column count unresolvable offline). The
+capture shim (integration/) confirms the counts in a cargo environment.
+Witness data is synthetic (the arithmetic is shape-dependent only).
 
 Run on a GPU box:   python tools/proof_trace_bench.py [--cpu]
 """
@@ -26,40 +29,53 @@ import pywrap as oracle  # noqa: E402  (CPU baseline leg + input gen only)
 
 R = 21888242871839275222246405745257275088548364400416034343698204186575808495617
 
-# (phase, kind, log_n, count) — SURVEY.md §3a structural estimates.
-# step20: sync-step k=20 (15 advice + 2 lookup-advice cols, ext domain 2^22).
-# agg23: the sync-step-compressed aggregation stage (K=23, 1 advice col +
-#        lookup, ext domain 2^25). committee24: committee-update aggregation
-#        (K=24, ext 2^26).
+# (phase, kind, log_n, count) — counts DERIVED from the PSE create_proof
+# structure for the pinned configs (CALLCOUNTS.md; committee24 = estimate).
+# Stage order follows the prover: advice commits, lookup permuted commits,
+# permutation z + lookup product commits, random poly, per-column iFFT +
+# coset-FFT for the quotient, extended icoset + h commits, SHPLONK.
+# Consecutive same-basis commits batch (the fused batch MSM API).
 TRACES = {
+    # sync-step k=20: 19 advice cols (15 base + 2 lookup-advice + 2 spread),
+    # L=3 lookups, NZ=11 z-chunks, j=3 h pieces, ext domain 2^22.
     "step20": (20, 22, [
-        ("advice commit",      "msm",   20, 17),
-        ("lookup commits",     "msm",   20,  6),
-        ("permutation z",      "msm",   20,  6),
-        ("quotient iFFT",      "intt",  20, 30),
-        ("quotient coset-FFT", "coset", 22, 30),
-        ("quotient h commits", "msm",   22,  4),
-        ("coset iFFT",         "icoset", 22, 1),
+        ("advice commit",      "msm",   20, 19),
+        ("lookup A'/S'",       "msm",   20,  6),
+        ("permutation z",      "msm",   20, 11),
+        ("lookup Z",           "msm",   20,  3),
+        ("vanishing random",   "msm",   20,  1),
+        ("quotient iFFT",      "intt",  20, 40),
+        ("quotient coset-FFT", "coset", 22, 40),
+        ("h extended icoset",  "icoset", 22, 1),
+        ("quotient h commits", "msm",   20,  3),
         ("SHPLONK multiopen",  "msm",   20,  2),
     ]),
+    # sync-step-compressed aggregation (K=23): A=2, L=1, NZ=2, ext 2^25.
     "agg23": (23, 25, [
         ("advice commit",      "msm",   23,  2),
-        ("lookup commits",     "msm",   23,  3),
+        ("lookup A'/S'",       "msm",   23,  2),
         ("permutation z",      "msm",   23,  2),
+        ("lookup Z",           "msm",   23,  1),
+        ("vanishing random",   "msm",   23,  1),
         ("quotient iFFT",      "intt",  23,  8),
         ("quotient coset-FFT", "coset", 25,  8),
-        ("quotient h commits", "msm",   23,  4),
-        ("coset iFFT",         "icoset", 25, 1),
+        ("h extended icoset",  "icoset", 25, 1),
+        ("quotient h commits", "msm",   23,  3),
         ("SHPLONK multiopen",  "msm",   23,  2),
     ]),
+    # committee-update aggregation (K=24): same formulas at 2^24/2^26.
+    # (The committee-update k=20 leaf circuit itself is NOT modeled: its
+    # zkevm-hashes SHA column count is unresolvable offline, CALLCOUNTS.md.)
     "committee24": (24, 26, [
         ("advice commit",      "msm",   24,  2),
-        ("lookup commits",     "msm",   24,  3),
+        ("lookup A'/S'",       "msm",   24,  2),
         ("permutation z",      "msm",   24,  2),
+        ("lookup Z",           "msm",   24,  1),
+        ("vanishing random",   "msm",   24,  1),
         ("quotient iFFT",      "intt",  24,  8),
         ("quotient coset-FFT", "coset", 26,  8),
-        ("quotient h commits", "msm",   24,  4),
-        ("coset iFFT",         "icoset", 26, 1),
+        ("h extended icoset",  "icoset", 26, 1),
+        ("quotient h commits", "msm",   24,  3),
         ("SHPLONK multiopen",  "msm",   24,  2),
     ]),
 }
@@ -164,7 +180,9 @@ def main():
         print(f"  {phase:22s} {kind:6s} 2^{log_n} x{count:3d}: {dt * 1e3:9.2f} ms")
     print(f"GPU hot-path total for one {args.config} proof: {total * 1e3:.1f} ms")
 
-    out = {"trace": f"{args.config} (SURVEY §3a estimates)",
+    prov = ("derived counts, CALLCOUNTS.md" if args.config != "committee24"
+            else "estimate (zkevm SHA cols unknown), CALLCOUNTS.md")
+    out = {"trace": f"{args.config} ({prov})",
            "gpu_hotpath_ms": round(total * 1e3, 1)}
 
     if args.cpu:
