@@ -103,6 +103,21 @@ int spectre_gpu_msm_g1_shard_device_timed(spectre_gpu_ctx*, int dev,
                                           uint32_t flags,
                                           uint8_t* out_partials,
                                           double out_ms[8]);
+/* Pipelined (async) shard MSM: enqueue the full pipeline on one of two
+ * per-device slots (own stream + scratch) and return WITHOUT synchronizing;
+ * *out_slot receives the slot id. Back-to-back MSMs on alternating slots
+ * overlap call i's latency-bound reduction tail with call i+1's
+ * digits/sort/accumulate (create_proof's ~45 commits per proof arrive
+ * exactly this way). out_partials (NUM_WINDOWS * 96 B) must stay valid
+ * until spectre_gpu_msm_slot_wait(ctx, dev, slot) returns. At most one
+ * call may be in flight per slot. */
+int spectre_gpu_msm_g1_shard_device_async(spectre_gpu_ctx*, int dev,
+                                          const void* d_bases,
+                                          const void* d_scalars, uint64_t n,
+                                          uint32_t flags,
+                                          uint8_t* out_partials,
+                                          int* out_slot);
+int spectre_gpu_msm_slot_wait(spectre_gpu_ctx*, int dev, int slot);
 /* Combine gathered shard partials (host-only, deterministic rank order):
  * partials = nshards * NUM_WINDOWS * 96 B. */
 int spectre_gpu_msm_g1_combine(const uint8_t* partials, uint32_t nshards,

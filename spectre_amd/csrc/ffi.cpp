@@ -60,14 +60,20 @@ void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
     if (!ctx) return;
     for (auto& ds : ctx->devs) {
         (void)hipSetDevice(ds.device_id);
-        for (void* p :
-             {(void*)ds.d_keys_in, (void*)ds.d_keys_out, (void*)ds.d_vals_in,
-              (void*)ds.d_vals_out, (void*)ds.d_sort_tmp, (void*)ds.d_offsets,
-              (void*)ds.d_buckets, (void*)ds.d_firstK, (void*)ds.d_lastK,
-              (void*)ds.d_firstP, (void*)ds.d_lastP,
-              (void*)ds.d_red, (void*)ds.d_scalars,
-              (void*)ds.d_bases, (void*)ds.d_ntt_tmp, (void*)ds.d_ntt_io,
-              (void*)ds.d_cosetA})
+        for (auto& sl : ds.slots) {
+            for (void* p : {(void*)sl.d_keys_in, (void*)sl.d_keys_out,
+                            (void*)sl.d_vals_in, (void*)sl.d_vals_out,
+                            (void*)sl.d_sort_tmp, (void*)sl.d_offsets,
+                            (void*)sl.d_buckets, (void*)sl.d_firstK,
+                            (void*)sl.d_lastK, (void*)sl.d_firstP,
+                            (void*)sl.d_lastP, (void*)sl.d_red})
+                if (p) (void)hipFree(p);
+            if (sl.stream && sl.stream != ds.stream)
+                (void)hipStreamDestroy(sl.stream);
+        }
+        for (void* p : {(void*)ds.d_scalars, (void*)ds.d_bases,
+                        (void*)ds.d_ntt_tmp, (void*)ds.d_ntt_io,
+                        (void*)ds.d_cosetA})
             if (p) (void)hipFree(p);
         for (int i = 0; i < 2; i++) {
             if (ds.h_stage[i]) (void)hipHostFree(ds.h_stage[i]);
@@ -232,6 +238,39 @@ int spectre_gpu_msm_g1_shard_device_timed(spectre_gpu_ctx* ctx, int dev,
     return msm_shard_device(ctx, dev, (const g1_affine*)d_bases,
                             (const uint8_t*)d_scalars, n, flags,
                             (g1_jac*)out_partials, out_ms);
+}
+
+int spectre_gpu_msm_g1_shard_device_async(spectre_gpu_ctx* ctx, int dev,
+                                          const void* d_bases,
+                                          const void* d_scalars, uint64_t n,
+                                          uint32_t flags,
+                                          uint8_t* out_partials,
+                                          int* out_slot) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    DeviceState& ds = ctx->devs[dev];
+    const int slot = ds.next_slot;
+    ds.next_slot ^= 1;
+    *out_slot = slot;
+    return msm_batch_shard_device(ctx, dev, (const g1_affine*)d_bases,
+                                  (const uint8_t*)d_scalars, 1, n, flags,
+                                  (g1_jac*)out_partials, nullptr,
+                                  /*sync=*/false, slot);
+}
+
+int spectre_gpu_msm_slot_wait(spectre_gpu_ctx* ctx, int dev, int slot) {
+    if (check_dev(ctx, dev)) return -1;
+    if (slot < 0 || slot > 1) {
+        set_err("slot_wait: slot %d out of range", slot);
+        return -1;
+    }
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    DeviceState& ds = ctx->devs[dev];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    if (ds.slots[slot].stream)
+        HIP_TRY(hipStreamSynchronize(ds.slots[slot].stream));
+    HIP_TRY(hipGetLastError());
+    return 0;
 }
 
 int spectre_gpu_msm_g1_device(spectre_gpu_ctx* ctx, int dev,

@@ -40,10 +40,13 @@ struct CachedBases {
     uint64_t n = 0;
 };
 
-struct DeviceState {
-    int device_id = 0;
-    hipStream_t stream = nullptr;
-    // ---- MSM scratch (grown on demand) ----
+// Per-slot MSM pipeline state: two slots per device let call i+1's
+// digits/sort/accumulate fill the machine while call i's latency-bound
+// reduction tail (~1.3 ms at <=1 wave/SIMD) drains on the other stream —
+// back-to-back commits (create_proof issues ~45 per proof) overlap instead
+// of serializing on the tail.
+struct MsmSlot {
+    hipStream_t stream = nullptr;  // slot 0 aliases DeviceState::stream
     uint32_t* d_keys_in = nullptr;
     uint32_t* d_keys_out = nullptr;
     uint32_t* d_vals_in = nullptr;
@@ -59,6 +62,13 @@ struct DeviceState {
     g1_jac* d_firstP = nullptr;
     g1_jac* d_lastP = nullptr;
     g1_jac* d_red = nullptr;        // reduction ping-pong (NB_TOTAL/CHUNK * 2)
+};
+
+struct DeviceState {
+    int device_id = 0;
+    hipStream_t stream = nullptr;
+    MsmSlot slots[2];
+    int next_slot = 0;  // round-robin for the async API
     uint8_t* d_scalars = nullptr;
     size_t scal_cap = 0;  // bytes
     g1_affine* d_bases = nullptr;
@@ -109,14 +119,16 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
                      g1_jac* winsums_host, double* stage_ms = nullptr);
 // batch variant: nbatch scalar vectors (batch-major, nbatch*n*32 B) over one
 // shared base set; winsums_host receives nbatch*MSM_NWIN Jacobian sums.
-// sync=false: enqueue the pipeline + async D2H of winsums on the device's
-// stream and return without synchronizing (multi-device overlap; the caller
-// must hipStreamSynchronize each stream before reading winsums_host).
+// sync=false: enqueue the pipeline + async D2H of winsums on the slot's
+// stream and return without synchronizing (multi-device / pipelined
+// overlap; the caller must hipStreamSynchronize that stream before reading
+// winsums_host). slot selects the per-device pipeline slot (scratch +
+// stream); slot 1's stream is created lazily.
 int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            const g1_affine* d_bases, const uint8_t* d_scalars,
                            uint32_t nbatch, uint64_t n, uint32_t flags,
                            g1_jac* winsums_host, double* stage_ms = nullptr,
-                           bool sync = true);
+                           bool sync = true, int slot = 0);
 
 // ntt.hip — pointwise Fr vector op on device buffers (synchronizes).
 int fr_vec_op_device(spectre_gpu_ctx* ctx, int dev, int op, const fp256* d_a,

@@ -260,7 +260,8 @@ __global__ __launch_bounds__(WSUM_THREADS, 2) void k_window_sum(
 }
 
 // ---- host orchestration ---------------------------------------------------
-static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
+static int ensure_msm_scratch(MsmSlot& ds, hipStream_t stream, uint64_t n,
+                              uint32_t nbatch) {
     const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
     const uint64_t nbt = (uint64_t)nbatch * MSM_NB_TOTAL;
     if (ds.ent_cap < ent) {
@@ -299,7 +300,7 @@ static int ensure_msm_scratch(DeviceState& ds, uint64_t n, uint32_t nbatch) {
     (void)hipcub::DeviceRadixSort::SortPairs(nullptr, sort_need, ds.d_keys_in,
                                              ds.d_keys_out, ds.d_vals_in,
                                              ds.d_vals_out, (int64_t)ent, 0,
-                                             32, ds.stream);
+                                             32, stream);
     if (ds.sort_tmp_cap < sort_need) {
         if (ds.d_sort_tmp) (void)hipFree(ds.d_sort_tmp);
         HIP_TRY(hipMalloc(&ds.d_sort_tmp, sort_need));
@@ -317,9 +318,19 @@ static int end_bit_for(uint64_t max_key) {
 int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            const g1_affine* d_bases, const uint8_t* d_scalars,
                            uint32_t nbatch, uint64_t n, uint32_t flags,
-                           g1_jac* winsums_host, double* stage_ms, bool sync) {
-    DeviceState& ds = ctx->devs[dev];
-    HIP_TRY(hipSetDevice(ds.device_id));
+                           g1_jac* winsums_host, double* stage_ms, bool sync,
+                           int slot) {
+    DeviceState& dstate = ctx->devs[dev];
+    HIP_TRY(hipSetDevice(dstate.device_id));
+    if (slot < 0 || slot > 1) {
+        set_err("msm: slot %d out of range", slot);
+        return -3;
+    }
+    MsmSlot& ds = dstate.slots[slot];
+    if (!ds.stream) {
+        if (slot == 0) ds.stream = dstate.stream;
+        else HIP_TRY(hipStreamCreate(&ds.stream));
+    }
     if (nbatch == 0 || nbatch > SPECTRE_MSM_MAX_BATCH) {
         set_err("msm: nbatch %u out of range [1,%d]", nbatch,
                 SPECTRE_MSM_MAX_BATCH);
@@ -340,7 +351,7 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
             g1j_set_inf(winsums_host[w]);
         return 0;
     }
-    int rc = ensure_msm_scratch(ds, n, nbatch);
+    int rc = ensure_msm_scratch(ds, ds.stream, n, nbatch);
     if (rc) return rc;
     const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
     const uint32_t nbt = nbatch * MSM_NB_TOTAL;

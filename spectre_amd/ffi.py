@@ -73,6 +73,13 @@ def load_library() -> ctypes.CDLL:
         c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
         c.c_void_p, c.POINTER(c.c_double),
     ]
+    lib.spectre_gpu_msm_g1_shard_device_async.restype = c.c_int
+    lib.spectre_gpu_msm_g1_shard_device_async.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
+        c.c_void_p, c.POINTER(c.c_int),
+    ]
+    lib.spectre_gpu_msm_slot_wait.restype = c.c_int
+    lib.spectre_gpu_msm_slot_wait.argtypes = [c.c_void_p, c.c_int, c.c_int]
     lib.spectre_gpu_msm_g1_combine.restype = c.c_int
     lib.spectre_gpu_msm_g1_combine.argtypes = [c.c_void_p, c.c_uint32, c.c_void_p]
     lib.spectre_gpu_ntt_fr.restype = c.c_int
@@ -229,6 +236,26 @@ class SpectreGpu:
             SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out)
         self._check(rc, "msm_g1_shard_device")
         return bytes(out)
+
+    def msm_shard_device_async(self, d_bases: int, d_scalars: int, n: int,
+                               canonical: bool = True, dev: int = 0):
+        """Enqueue a full shard MSM on one of two per-device pipeline slots
+        and return (partials_buffer, slot) WITHOUT synchronizing. Call
+        msm_slot_wait(slot) before reading the buffer; keep the returned
+        ctypes buffer referenced until then (async D2H writes into it)."""
+        out = (ctypes.c_uint8 * PARTIALS_BYTES)()
+        slot = ctypes.c_int(-1)
+        rc = self._lib.spectre_gpu_msm_g1_shard_device_async(
+            self._ctx, dev, ctypes.c_void_p(d_bases),
+            ctypes.c_void_p(d_scalars), n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY, out,
+            ctypes.byref(slot))
+        self._check(rc, "msm_g1_shard_device_async")
+        return out, slot.value
+
+    def msm_slot_wait(self, slot: int, dev: int = 0) -> None:
+        rc = self._lib.spectre_gpu_msm_slot_wait(self._ctx, dev, slot)
+        self._check(rc, "msm_slot_wait")
 
     def msm_shard_device_timed(self, d_bases: int, d_scalars: int, n: int,
                                canonical: bool = True, dev: int = 0):

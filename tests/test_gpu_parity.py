@@ -107,6 +107,41 @@ def test_msm_bases_cache(gpu, oracle):
     assert r1 == r2 == oracle.msm(bs, sc, n)
 
 
+def test_msm_async_pipeline_matches_sync(gpu, oracle):
+    """Depth-2 pipelined MSMs on the two per-device slots: every in-flight
+    call returns exactly the synchronous result, including when two
+    different inputs are in flight on the two slots at once."""
+    from spectre_amd import ffi
+    n = 1 << 14
+    sc_a, bs = oracle.gen_msm_inputs(n, 61, fast=True)
+    sc_b = oracle.gen_msm_inputs(n, 62, fast=True)[0]
+    d_b = gpu.malloc(64 * n)
+    d_sa = gpu.malloc(32 * n)
+    d_sb = gpu.malloc(32 * n)
+    gpu.upload(d_b, bs)
+    gpu.upload(d_sa, sc_a)
+    gpu.upload(d_sb, sc_b)
+    want_a = ffi.combine_partials(gpu.msm_shard_device(d_b, d_sa, n), 1)
+    want_b = ffi.combine_partials(gpu.msm_shard_device(d_b, d_sb, n), 1)
+    # interleave: a,b,a,b,... with both slots busy simultaneously
+    pend = []
+    got = []
+    for i in range(6):
+        buf, slot = gpu.msm_shard_device_async(d_b, d_sa if i % 2 == 0 else d_sb, n)
+        pend.append((buf, slot))
+        if len(pend) == 2:
+            b0, s0 = pend.pop(0)
+            gpu.msm_slot_wait(s0)
+            got.append(ffi.combine_partials(bytes(b0), 1))
+    for b0, s0 in pend:
+        gpu.msm_slot_wait(s0)
+        got.append(ffi.combine_partials(bytes(b0), 1))
+    for i, g in enumerate(got):
+        assert g == (want_a if i % 2 == 0 else want_b), f"call {i}"
+    for p in (d_b, d_sa, d_sb):
+        gpu.free(p)
+
+
 def test_msm_shard_combine_matches_direct(gpu, oracle):
     """Two shards on one device + host combine == unsharded result — the
     exact exchange the multi-GPU path performs (bit-identical by affine
