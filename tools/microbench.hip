@@ -53,10 +53,47 @@ __global__ __launch_bounds__(256, 2) void k_mul_chain(const g1_affine* pts,
     out[t].X = a;
 }
 
+#include "../spectre_amd/csrc/ff_asm.hpp"
+
+// asm column-Montgomery multiply chain (same shape as k_mul_chain)
+__global__ __launch_bounds__(256, 2) void k_mul_cols_chain(
+    const g1_affine* pts, g1_jac* out, int nwork) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= nwork) return;
+    fp256 a = pts[t % 1024].x, b = pts[t % 1024].y;
+    for (int i = 0; i < ITERS * 16; i++) ff_mul_cols<Fq>(a, a, b);
+    out[t].X = a;
+}
+
+// correctness: ff_mul_cols must equal ff_mul bit-for-bit on pseudorandom
+// reduced inputs (both Fq and Fr), chained so errors compound and surface.
+__global__ void k_mul_cols_check(uint32_t* bad, int nwork) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= nwork) return;
+    uint64_t s = 0x9e3779b97f4a7c15ull * (t + 1);
+    fp256 a, b;
+    for (int i = 0; i < 8; i++) {
+        s = s * 6364136223846793005ull + 1442695040888963407ull;
+        a.l[i] = (uint32_t)(s >> 32);
+        s = s * 6364136223846793005ull + 1442695040888963407ull;
+        b.l[i] = (uint32_t)(s >> 32);
+    }
+    a.l[7] &= 0x0fffffffu;  // < 2^252 < p (both moduli)
+    b.l[7] &= 0x0fffffffu;
+    fp256 xq = a, yq = a, xr = b, yr = b;
+    for (int i = 0; i < 64; i++) {
+        ff_mul<Fq>(xq, xq, b);
+        ff_mul_cols<Fq>(yq, yq, b);
+        ff_mul<Fr>(xr, xr, a);
+        ff_mul_cols<Fr>(yr, yr, a);
+    }
+    if (!ff_eq(xq, yq) || !ff_eq(xr, yr)) atomicAdd(bad, 1u);
+}
+
 static double time_kernel(void (*fn)(const g1_affine*, g1_jac*, int),
                           const g1_affine* pts, g1_jac* out, int nwork,
                           int blocks) {
-    (void)hipEvent_t e0, e1;
+    hipEvent_t e0, e1;
     (void)hipEventCreate(&e0);
     (void)hipEventCreate(&e1);
     hipLaunchKernelGGL(fn, dim3(blocks), dim3(256), 0, 0, pts, out, nwork);
@@ -95,9 +132,24 @@ int main() {
     double msm = time_kernel(k_mul_chain, pts, out, nwork, nwork / 256);
     printf("ff_mul chain: %.3f ms  (%.1f M mul/s)\n", msm,
            (double)nwork * ITERS * 16 / msm / 1e3);
+    {
+        uint32_t* bad;
+        (void)hipMalloc(&bad, 4);
+        (void)hipMemset(bad, 0, 4);
+        hipLaunchKernelGGL(k_mul_cols_check, dim3(1024), dim3(256), 0, 0, bad,
+                           1024 * 256);
+        uint32_t h_bad = 1;
+        (void)hipMemcpy(&h_bad, bad, 4, hipMemcpyDeviceToHost);
+        printf("ff_mul_cols parity vs ff_mul (256K lanes x 64 chained, "
+               "Fq+Fr): %s (%u bad)\n", h_bad ? "FAIL" : "ok", h_bad);
+        double msc = time_kernel(k_mul_cols_chain, pts, out, nwork, nwork / 256);
+        printf("ff_mul_cols chain: %.3f ms  (%.1f M mul/s, %.2fx vs C)\n",
+               msc, (double)nwork * ITERS * 16 / msc / 1e3, msm / msc);
+        (void)hipFree(bad);
+    }
     // raw mad64 rate
     {
-        (void)hipEvent_t e0, e1;
+        hipEvent_t e0, e1;
         (void)hipEventCreate(&e0);
         (void)hipEventCreate(&e1);
         uint32_t iters = 4096;
