@@ -68,6 +68,7 @@ void spectre_gpu_destroy(spectre_gpu_ctx* ctx) {
                             (void*)sl.d_lastK, (void*)sl.d_firstP,
                             (void*)sl.d_lastP, (void*)sl.d_red})
                 if (p) (void)hipFree(p);
+            if (sl.h_wins) (void)hipHostFree(sl.h_wins);
             if (sl.stream && sl.stream != ds.stream)
                 (void)hipStreamDestroy(sl.stream);
         }
@@ -265,12 +266,7 @@ int spectre_gpu_msm_slot_wait(spectre_gpu_ctx* ctx, int dev, int slot) {
         return -1;
     }
     std::lock_guard<std::recursive_mutex> lk(ctx->mu);
-    DeviceState& ds = ctx->devs[dev];
-    HIP_TRY(hipSetDevice(ds.device_id));
-    if (ds.slots[slot].stream)
-        HIP_TRY(hipStreamSynchronize(ds.slots[slot].stream));
-    HIP_TRY(hipGetLastError());
-    return 0;
+    return msm_slot_drain(ctx, dev, slot);
 }
 
 int spectre_gpu_msm_g1_device(spectre_gpu_ctx* ctx, int dev,
@@ -471,8 +467,8 @@ int spectre_gpu_msm_g1(spectre_gpu_ctx* ctx, uint64_t bases_id,
         }
     }
     for (int d = 0; d < num_gpus; d++) {
-        HIP_TRY(hipSetDevice(ctx->devs[d].device_id));
-        HIP_TRY(hipStreamSynchronize(ctx->devs[d].stream));
+        int rc = msm_slot_drain(ctx, d, 0);
+        if (rc) return rc;
     }
     return spectre_gpu_msm_g1_combine((const uint8_t*)wins.data(),
                                       (uint32_t)num_gpus, out_affine);

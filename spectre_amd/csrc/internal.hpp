@@ -62,6 +62,14 @@ struct MsmSlot {
     g1_jac* d_firstP = nullptr;
     g1_jac* d_lastP = nullptr;
     g1_jac* d_red = nullptr;        // reduction ping-pong (NB_TOTAL/CHUNK * 2)
+    // Window sums come back through a PINNED per-slot buffer: an async D2H
+    // into caller (pageable) memory silently blocks the calling thread on
+    // ROCm, which serialized the whole two-slot pipeline in r2's first
+    // measurement. The wait/drain step memcpys pinned -> caller.
+    g1_jac* h_wins = nullptr;   // pinned, h_wins_cap elements
+    size_t h_wins_cap = 0;
+    g1_jac* pending_dst = nullptr;  // caller buffer for the in-flight call
+    uint32_t pending_n = 0;         // elements pending (nbatch * NWIN)
 };
 
 struct DeviceState {
@@ -129,6 +137,9 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            uint32_t nbatch, uint64_t n, uint32_t flags,
                            g1_jac* winsums_host, double* stage_ms = nullptr,
                            bool sync = true, int slot = 0);
+
+// msm.hip — sync a slot's stream and deliver pending window sums.
+int msm_slot_drain(spectre_gpu_ctx* ctx, int dev, int slot);
 
 // ntt.hip — pointwise Fr vector op on device buffers (synchronizes).
 int fr_vec_op_device(spectre_gpu_ctx* ctx, int dev, int op, const fp256* d_a,

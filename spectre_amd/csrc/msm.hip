@@ -400,8 +400,14 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
     hipLaunchKernelGGL(k_window_sum, dim3(nbatch * MSM_NWIN),
                        dim3(WSUM_THREADS), 0, st, red0, red1);
     STAMP(6);
-    HIP_TRY(hipMemcpyAsync(winsums_host, red1,
-                           (size_t)nbatch * MSM_NWIN * sizeof(g1_jac),
+    const uint32_t nw = nbatch * MSM_NWIN;
+    if (ds.h_wins_cap < nw) {
+        if (ds.h_wins) (void)hipHostFree(ds.h_wins);
+        ds.h_wins_cap = 0;
+        HIP_TRY(hipHostMalloc((void**)&ds.h_wins, (size_t)nw * sizeof(g1_jac)));
+        ds.h_wins_cap = nw;
+    }
+    HIP_TRY(hipMemcpyAsync(ds.h_wins, red1, (size_t)nw * sizeof(g1_jac),
                            hipMemcpyDeviceToHost, st));
     uint32_t ent_real = 0;
     if (stage_ms)
@@ -409,10 +415,13 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                                hipMemcpyDeviceToHost, st));
     if (!sync && !stage_ms) {
         HIP_TRY(hipGetLastError());
-        return 0;  // caller synchronizes the stream before reading winsums
+        ds.pending_dst = winsums_host;  // drained by msm_slot_drain
+        ds.pending_n = nw;
+        return 0;
     }
     HIP_TRY(hipStreamSynchronize(st));
     HIP_TRY(hipGetLastError());
+    memcpy(winsums_host, ds.h_wins, (size_t)nw * sizeof(g1_jac));
     if (stage_ms) {
         float ms;
         for (int i = 0; i < 6; i++) {
@@ -433,4 +442,24 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
                      g1_jac* winsums_host, double* stage_ms) {
     return msm_batch_shard_device(ctx, dev, d_bases, d_scalars, 1, n, flags,
                                   winsums_host, stage_ms);
+}
+
+// Synchronize a slot's stream and deliver the pending window sums (pinned ->
+// caller memory). Idempotent when nothing is pending.
+int msm_slot_drain(spectre_gpu_ctx* ctx, int dev, int slot) {
+    DeviceState& dstate = ctx->devs[dev];
+    if (slot < 0 || slot > 1) {
+        set_err("slot_drain: slot %d out of range", slot);
+        return -1;
+    }
+    MsmSlot& sl = dstate.slots[slot];
+    HIP_TRY(hipSetDevice(dstate.device_id));
+    if (sl.stream) HIP_TRY(hipStreamSynchronize(sl.stream));
+    HIP_TRY(hipGetLastError());
+    if (sl.pending_dst) {
+        memcpy(sl.pending_dst, sl.h_wins, (size_t)sl.pending_n * sizeof(g1_jac));
+        sl.pending_dst = nullptr;
+        sl.pending_n = 0;
+    }
+    return 0;
 }
