@@ -29,6 +29,12 @@
 //
 // All Fr math is 8x32-limb Montgomery (ff.hpp); data stays in Montgomery form
 // end-to-end exactly as halo2 holds its &[Fr] slices.
+//
+// This TU keeps the C CIOS multiply: the asm column multiply regressed the
+// 1024-thread LDS kernels ~7% (2^23 fwd 2.59 -> 2.77 ms — register pressure
+// at the 4-waves/SIMD occupancy these kernels need), while it wins ~28% in
+// the 256-thread MSM kernels. Per-TU choice, measured r2.
+#define SPECTRE_NO_ASM_MUL 1
 #include "internal.hpp"
 
 #define THREADS 256       // pow-table builder
