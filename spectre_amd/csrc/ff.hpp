@@ -187,6 +187,13 @@ template <class C> FF_HD void ff_mul_cios(fp256& o, const fp256& a, const fp256&
 // Single product: acc += a*b (64-bit mad), ovf += carry-out. Explicit
 // carry-mask pairs ("=s"): a vcc clobber makes hipcc pad every block
 // boundary with s_nop hazards.
+// MEASURED NEGATIVE RESULT (r2): fusing 2/4 products per asm block to cut
+// those boundary s_nops produces WRONG results at full-range operands — a
+// v_mad_u64_u32 reading the 64-bit acc written by a mad one instruction
+// earlier needs >= 2 intervening issue slots (the compiler's conservative
+// inter-block s_nop was supplying exactly that); and the s_nops are hidden
+// by co-resident waves anyway (grouped rate unchanged at 135 G/s). Keep
+// one mad+addc pair per block.
 __device__ __forceinline__ void ff_mad64_(uint64_t& acc, uint32_t& ovf,
                                           uint32_t a, uint32_t b) {
     uint64_t c;
@@ -203,102 +210,6 @@ __device__ __forceinline__ void ff_mad64_s_(uint64_t& acc, uint32_t& ovf,
                  : "+v"(acc), "+v"(ovf), "=s"(c)
                  : "v"(a), "s"(b_uniform));
 }
-// Fused 2- and 4-product blocks: fewer asm-block boundaries = fewer
-// compiler-inserted s_nops between blocks (~1 per boundary). Alternating
-// carry pairs c0/c1 keep consecutive sdst writes off the same sgprs.
-__device__ __forceinline__ void ff_mad64_x2(uint64_t& acc, uint32_t& ovf,
-                                            uint32_t a0, uint32_t b0,
-                                            uint32_t a1, uint32_t b1) {
-    uint64_t c0, c1;
-    asm volatile("v_mad_u64_u32 %0, %2, %4, %5, %0\n\t"
-                 "v_addc_co_u32 %1, %2, 0, %1, %2\n\t"
-                 "v_mad_u64_u32 %0, %3, %6, %7, %0\n\t"
-                 "v_addc_co_u32 %1, %3, 0, %1, %3"
-                 : "+v"(acc), "+v"(ovf), "=s"(c0), "=s"(c1)
-                 : "v"(a0), "v"(b0), "v"(a1), "v"(b1));
-}
-__device__ __forceinline__ void ff_mad64_s_x2(uint64_t& acc, uint32_t& ovf,
-                                              uint32_t a0, uint32_t b0,
-                                              uint32_t a1, uint32_t b1) {
-    uint64_t c0, c1;
-    asm volatile("v_mad_u64_u32 %0, %2, %4, %5, %0\n\t"
-                 "v_addc_co_u32 %1, %2, 0, %1, %2\n\t"
-                 "v_mad_u64_u32 %0, %3, %6, %7, %0\n\t"
-                 "v_addc_co_u32 %1, %3, 0, %1, %3"
-                 : "+v"(acc), "+v"(ovf), "=s"(c0), "=s"(c1)
-                 : "v"(a0), "s"(b0), "v"(a1), "s"(b1));
-}
-__device__ __forceinline__ void ff_mad64_x4(uint64_t& acc, uint32_t& ovf,
-                                            uint32_t a0, uint32_t b0,
-                                            uint32_t a1, uint32_t b1,
-                                            uint32_t a2, uint32_t b2,
-                                            uint32_t a3, uint32_t b3) {
-    uint64_t c0, c1;
-    asm volatile("v_mad_u64_u32 %0, %2, %4, %5, %0\n\t"
-                 "v_addc_co_u32 %1, %2, 0, %1, %2\n\t"
-                 "v_mad_u64_u32 %0, %3, %6, %7, %0\n\t"
-                 "v_addc_co_u32 %1, %3, 0, %1, %3\n\t"
-                 "v_mad_u64_u32 %0, %2, %8, %9, %0\n\t"
-                 "v_addc_co_u32 %1, %2, 0, %1, %2\n\t"
-                 "v_mad_u64_u32 %0, %3, %10, %11, %0\n\t"
-                 "v_addc_co_u32 %1, %3, 0, %1, %3"
-                 : "+v"(acc), "+v"(ovf), "=s"(c0), "=s"(c1)
-                 : "v"(a0), "v"(b0), "v"(a1), "v"(b1), "v"(a2), "v"(b2),
-                   "v"(a3), "v"(b3));
-}
-__device__ __forceinline__ void ff_mad64_s_x4(uint64_t& acc, uint32_t& ovf,
-                                              uint32_t a0, uint32_t b0,
-                                              uint32_t a1, uint32_t b1,
-                                              uint32_t a2, uint32_t b2,
-                                              uint32_t a3, uint32_t b3) {
-    uint64_t c0, c1;
-    asm volatile("v_mad_u64_u32 %0, %2, %4, %5, %0\n\t"
-                 "v_addc_co_u32 %1, %2, 0, %1, %2\n\t"
-                 "v_mad_u64_u32 %0, %3, %6, %7, %0\n\t"
-                 "v_addc_co_u32 %1, %3, 0, %1, %3\n\t"
-                 "v_mad_u64_u32 %0, %2, %8, %9, %0\n\t"
-                 "v_addc_co_u32 %1, %2, 0, %1, %2\n\t"
-                 "v_mad_u64_u32 %0, %3, %10, %11, %0\n\t"
-                 "v_addc_co_u32 %1, %3, 0, %1, %3"
-                 : "+v"(acc), "+v"(ovf), "=s"(c0), "=s"(c1)
-                 : "v"(a0), "s"(b0), "v"(a1), "s"(b1), "v"(a2), "s"(b2),
-                   "v"(a3), "s"(b3));
-}
-// run of products av[i] * (VPART ? bv[K-i] : mod(K-i)) for i in [LO, HI),
-// grouped into 4/2/1-product asm blocks; K/LO/HI compile-time so every
-// C::mod() resolves to an immediate (template recursion guarantees it).
-template <class C, int K, int LO, int HI, int VPART>
-struct FfColRun {
-    static __device__ __forceinline__ void run(uint64_t& acc, uint32_t& ovf,
-                                               const uint32_t* av,
-                                               const uint32_t* bv) {
-        if constexpr (HI - LO >= 4) {
-            if constexpr (VPART)
-                ff_mad64_x4(acc, ovf, av[LO], bv[K - LO], av[LO + 1],
-                            bv[K - LO - 1], av[LO + 2], bv[K - LO - 2],
-                            av[LO + 3], bv[K - LO - 3]);
-            else
-                ff_mad64_s_x4(acc, ovf, av[LO], C::mod(K - LO), av[LO + 1],
-                              C::mod(K - LO - 1), av[LO + 2],
-                              C::mod(K - LO - 2), av[LO + 3],
-                              C::mod(K - LO - 3));
-            FfColRun<C, K, LO + 4, HI, VPART>::run(acc, ovf, av, bv);
-        } else if constexpr (HI - LO >= 2) {
-            if constexpr (VPART)
-                ff_mad64_x2(acc, ovf, av[LO], bv[K - LO], av[LO + 1],
-                            bv[K - LO - 1]);
-            else
-                ff_mad64_s_x2(acc, ovf, av[LO], C::mod(K - LO), av[LO + 1],
-                              C::mod(K - LO - 1));
-            FfColRun<C, K, LO + 2, HI, VPART>::run(acc, ovf, av, bv);
-        } else if constexpr (HI - LO == 1) {
-            if constexpr (VPART)
-                ff_mad64_(acc, ovf, av[LO], bv[K - LO]);
-            else
-                ff_mad64_s_(acc, ovf, av[LO], C::mod(K - LO));
-        }
-    }
-};
 template <class C, int K>
 __device__ __forceinline__ void detail_ff_low_cols(uint64_t& acc,
                                                    uint32_t& ovf,
@@ -306,8 +217,10 @@ __device__ __forceinline__ void detail_ff_low_cols(uint64_t& acc,
                                                    const uint32_t* b,
                                                    uint32_t* m) {
     if constexpr (K < 8) {
-        FfColRun<C, K, 0, K + 1, 1>::run(acc, ovf, a, b);      // a_i * b_{K-i}
-        FfColRun<C, K, 0, K, 0>::run(acc, ovf, m, nullptr);    // m_i * p_{K-i}
+#pragma unroll
+        for (int i = 0; i <= K; i++) ff_mad64_(acc, ovf, a[i], b[K - i]);
+#pragma unroll
+        for (int i = 0; i < K; i++) ff_mad64_s_(acc, ovf, m[i], C::mod(K - i));
         m[K] = (uint32_t)acc * C::inv();
         ff_mad64_s_(acc, ovf, m[K], C::mod(0));
         acc = (acc >> 32) | ((uint64_t)ovf << 32);
@@ -323,8 +236,10 @@ __device__ __forceinline__ void detail_ff_high_cols(uint64_t& acc,
                                                     const uint32_t* m,
                                                     uint32_t* out) {
     if constexpr (K < 15) {
-        FfColRun<C, K, K - 7, 8, 1>::run(acc, ovf, a, b);   // a_i * b_{K-i}
-        FfColRun<C, K, K - 7, 8, 0>::run(acc, ovf, m, nullptr);  // m_i * p
+#pragma unroll
+        for (int i = K - 7; i < 8; i++) ff_mad64_(acc, ovf, a[i], b[K - i]);
+#pragma unroll
+        for (int i = K - 7; i < 8; i++) ff_mad64_s_(acc, ovf, m[i], C::mod(K - i));
         out[K - 8] = (uint32_t)acc;
         acc = (acc >> 32) | ((uint64_t)ovf << 32);
         ovf = 0;
