@@ -293,6 +293,180 @@ static void g1j_mul(g1j* o, const g1a* p, const fe* k_canon) {
     *o = acc;
 }
 
+
+/* ------------------------------------------------------------ G2 (Fq2) */
+/* BN254 G2: y^2 = x^3 + b2 over Fq2 = Fq[u]/(u^2+1), b2 = 3/(9+u) (D-twist;
+ * halo2curves bn256::G2, SURVEY.md §8a minor row — sizes <= 2 in the
+ * reference: SRS/verifier-side algebra only, so the oracle covers it and no
+ * GPU kernel exists). Affine memory image (G2Affine): x.c0||x.c1||y.c0||y.c1,
+ * 32 B LE Montgomery each; identity = 128 zero bytes. Affine chord/tangent
+ * formulas with Fermat inversion (sizes are tiny). */
+typedef struct { fe c0, c1; } fq2;
+typedef struct { fq2 x, y; int inf; } g2a;
+
+static void fq2_add(fq2* o, const fq2* a, const fq2* b) {
+    f_add(&FQ, &o->c0, &a->c0, &b->c0);
+    f_add(&FQ, &o->c1, &a->c1, &b->c1);
+}
+static void fq2_sub(fq2* o, const fq2* a, const fq2* b) {
+    f_sub(&FQ, &o->c0, &a->c0, &b->c0);
+    f_sub(&FQ, &o->c1, &a->c1, &b->c1);
+}
+static void fq2_neg(fq2* o, const fq2* a) {
+    f_neg(&FQ, &o->c0, &a->c0);
+    f_neg(&FQ, &o->c1, &a->c1);
+}
+static void fq2_mul(fq2* o, const fq2* a, const fq2* b) {
+    fe t0, t1, t2, t3;
+    f_mul(&FQ, &t0, &a->c0, &b->c0);
+    f_mul(&FQ, &t1, &a->c1, &b->c1);
+    f_mul(&FQ, &t2, &a->c0, &b->c1);
+    f_mul(&FQ, &t3, &a->c1, &b->c0);
+    f_sub(&FQ, &o->c0, &t0, &t1);   /* u^2 = -1 */
+    f_add(&FQ, &o->c1, &t2, &t3);
+}
+static void fq2_inv(fq2* o, const fq2* a) {
+    /* (c0 + c1 u)^-1 = (c0 - c1 u) / (c0^2 + c1^2) */
+    fe n0, n1, d;
+    f_sqr(&FQ, &n0, &a->c0);
+    f_sqr(&FQ, &n1, &a->c1);
+    f_add(&FQ, &d, &n0, &n1);
+    f_inv(&FQ, &d, &d);
+    f_mul(&FQ, &o->c0, &a->c0, &d);
+    fe nc1;
+    f_neg(&FQ, &nc1, &a->c1);
+    f_mul(&FQ, &o->c1, &nc1, &d);
+}
+static int fq2_is_zero(const fq2* a) {
+    return fe_is_zero(&a->c0) && fe_is_zero(&a->c1);
+}
+
+static void g2a_from_bytes(g2a* p, const uint8_t b[128]) {
+    fe_from_bytes(&p->x.c0, b);
+    fe_from_bytes(&p->x.c1, b + 32);
+    fe_from_bytes(&p->y.c0, b + 64);
+    fe_from_bytes(&p->y.c1, b + 96);
+    p->inf = fq2_is_zero(&p->x) && fq2_is_zero(&p->y);
+}
+static void g2a_to_bytes(uint8_t b[128], const g2a* p) {
+    if (p->inf) { memset(b, 0, 128); return; }
+    fe_to_bytes(b, &p->x.c0);
+    fe_to_bytes(b + 32, &p->x.c1);
+    fe_to_bytes(b + 64, &p->y.c0);
+    fe_to_bytes(b + 96, &p->y.c1);
+}
+static void g2a_add(g2a* o, const g2a* a, const g2a* b) {
+    if (a->inf) { *o = *b; return; }
+    if (b->inf) { *o = *a; return; }
+    fq2 lam, t;
+    if (fe_eq(&a->x.c0, &b->x.c0) && fe_eq(&a->x.c1, &b->x.c1)) {
+        fq2 ysum;
+        fq2_add(&ysum, &a->y, &b->y);
+        if (fq2_is_zero(&ysum)) { memset(o, 0, sizeof *o); o->inf = 1; return; }
+        /* tangent: 3 x^2 / 2y */
+        fq2 num, den;
+        fq2_mul(&num, &a->x, &a->x);
+        fq2_add(&t, &num, &num);
+        fq2_add(&num, &t, &num);
+        fq2_add(&den, &a->y, &a->y);
+        fq2_inv(&den, &den);
+        fq2_mul(&lam, &num, &den);
+    } else {
+        fq2 num, den;
+        fq2_sub(&num, &b->y, &a->y);
+        fq2_sub(&den, &b->x, &a->x);
+        fq2_inv(&den, &den);
+        fq2_mul(&lam, &num, &den);
+    }
+    g2a r;
+    r.inf = 0;
+    fq2_mul(&r.x, &lam, &lam);
+    fq2_sub(&r.x, &r.x, &a->x);
+    fq2_sub(&r.x, &r.x, &b->x);
+    fq2_sub(&t, &a->x, &r.x);
+    fq2_mul(&r.y, &lam, &t);
+    fq2_sub(&r.y, &r.y, &a->y);
+    *o = r;
+}
+static void g2a_mul(g2a* o, const g2a* p, const fe* k_canon) {
+    g2a acc, base = *p;
+    memset(&acc, 0, sizeof acc);
+    acc.inf = 1;
+    for (int i = 0; i < 256; i++) {
+        if ((k_canon->l[i >> 6] >> (i & 63)) & 1) {
+            g2a t = acc;
+            g2a_add(&acc, &t, &base);
+        }
+        g2a t = base;
+        g2a_add(&base, &t, &t);
+    }
+    *o = acc;
+}
+
+void oracle_g2_add(const uint8_t a[128], const uint8_t b[128], uint8_t out[128]) {
+    g2a A, B, O;
+    g2a_from_bytes(&A, a);
+    g2a_from_bytes(&B, b);
+    g2a_add(&O, &A, &B);
+    g2a_to_bytes(out, &O);
+}
+void oracle_g2_neg(const uint8_t a[128], uint8_t out[128]) {
+    g2a A;
+    g2a_from_bytes(&A, a);
+    if (!A.inf) fq2_neg(&A.y, &A.y);
+    g2a_to_bytes(out, &A);
+}
+void oracle_g2_mul(const uint8_t p[128], const uint8_t k_canon[32], uint8_t out[128]) {
+    g2a A, O;
+    fe k;
+    g2a_from_bytes(&A, p);
+    fe_from_bytes(&k, k_canon);
+    g2a_mul(&O, &A, &k);
+    g2a_to_bytes(out, &O);
+}
+int oracle_g2_is_on_curve(const uint8_t p[128]) {
+    g2a A;
+    g2a_from_bytes(&A, p);
+    if (A.inf) return 1;
+    /* b2 = 3/(9+u), computed on the fly (values in Montgomery form) */
+    fq2 nine_u, b2, three, rhs, lhs, t;
+    nine_u.c0 = FQ.one;
+    f_add(&FQ, &nine_u.c0, &nine_u.c0, &FQ.one);
+    fe two = nine_u.c0;                       /* 2 */
+    f_add(&FQ, &nine_u.c0, &two, &two);       /* 4 */
+    f_add(&FQ, &nine_u.c0, &nine_u.c0, &nine_u.c0); /* 8 */
+    f_add(&FQ, &nine_u.c0, &nine_u.c0, &FQ.one);    /* 9 */
+    nine_u.c1 = FQ.one;
+    fq2_inv(&b2, &nine_u);
+    f_add(&FQ, &three.c0, &two, &FQ.one);
+    three.c1 = FE_ZERO;
+    fq2_mul(&b2, &b2, &three);
+    fq2_mul(&t, &A.x, &A.x);
+    fq2_mul(&rhs, &t, &A.x);
+    fq2_add(&rhs, &rhs, &b2);
+    fq2_mul(&lhs, &A.y, &A.y);
+    return fe_eq(&lhs.c0, &rhs.c0) && fe_eq(&lhs.c1, &rhs.c1);
+}
+/* Sum_i scalars_i * bases_i over G2 — the reference's only G2 MSMs are the
+ * size <= 2 SRS/verifier ones, so a simple serial sum is the
+ * whole requirement. */
+void oracle_msm_g2(const uint8_t* bases, const uint8_t* scalars_canon,
+                   uint64_t n, uint8_t out[128]) {
+    g2a acc;
+    memset(&acc, 0, sizeof acc);
+    acc.inf = 1;
+    for (uint64_t i = 0; i < n; i++) {
+        g2a p, t, prod;
+        fe k;
+        g2a_from_bytes(&p, bases + 128 * i);
+        fe_from_bytes(&k, scalars_canon + 32 * i);
+        g2a_mul(&prod, &p, &k);
+        t = acc;
+        g2a_add(&acc, &t, &prod);
+    }
+    g2a_to_bytes(out, &acc);
+}
+
 /* ------------------------------------------------------------ public: fields */
 #define FIELD_WRAP(name, ctx, op)                                              \
     void name(const uint8_t a[32], const uint8_t b[32], uint8_t out[32]) {     \

@@ -48,6 +48,16 @@ void oracle_g1_neg(const uint8_t a[64], uint8_t out[64]);
 void oracle_g1_mul(const uint8_t p[64], const uint8_t k_canon[32], uint8_t out[64]);
 int  oracle_g1_is_on_curve(const uint8_t p[64]);
 
+/* G2 (Fq2 twist; SURVEY §8a minor row — size <= 2, oracle-only, no kernel).
+ * Affine image: x.c0||x.c1||y.c0||y.c1 (32 B LE Montgomery each);
+ * identity = 128 zero bytes. Scalars canonical LE. */
+void oracle_g2_add(const uint8_t a[128], const uint8_t b[128], uint8_t out[128]);
+void oracle_g2_neg(const uint8_t a[128], uint8_t out[128]);
+void oracle_g2_mul(const uint8_t p[128], const uint8_t k_canon[32], uint8_t out[128]);
+int  oracle_g2_is_on_curve(const uint8_t p[128]);
+void oracle_msm_g2(const uint8_t* bases, const uint8_t* scalars_canon,
+                   uint64_t n, uint8_t out[128]);
+
 /* -- the hot path --------------------------------------------------------- */
 /* Sum_i scalars[i]*bases[i]; scalars 32B each (canonical if
  * scalars_canonical, else Montgomery); bases 64B affine; OpenMP Pippenger. */

@@ -62,6 +62,23 @@ def g1_neg(a): return _call1("oracle_g1_neg", a, 64)
 def g1_mul(p, k): return _call2("oracle_g1_mul", p, k, 64)
 
 
+def g2_add(a, b): return _call2("oracle_g2_add", a, b, 128)
+def g2_neg(a): return _call1("oracle_g2_neg", a, 128)
+def g2_mul(p, k): return _call2("oracle_g2_mul", p, k, 128)
+
+
+def g2_is_on_curve(p: bytes) -> bool:
+    l = lib()
+    l.oracle_g2_is_on_curve.restype = ctypes.c_int
+    return bool(l.oracle_g2_is_on_curve(_buf(p)))
+
+
+def msm_g2(bases: bytes, scalars_canon: bytes, n: int) -> bytes:
+    out = (ctypes.c_uint8 * 128)()
+    lib().oracle_msm_g2(_buf(bases), _buf(scalars_canon), ctypes.c_uint64(n), out)
+    return bytes(out)
+
+
 def g1_is_on_curve(p: bytes) -> bool:
     return bool(lib().oracle_g1_is_on_curve(_buf(p)))
 

@@ -174,3 +174,93 @@ def distribute_powers(a, g):
         out.append(x * cur % R)
         cur = cur * g % R
     return out
+
+
+# ---------------------------------------------------------------- G2 (Fq2)
+# BN254 G2: y^2 = x^3 + b2 over Fq2 = Fq[u]/(u^2 + 1), b2 = 3/(9+u) (the
+# D-type sextic twist; halo2curves bn256::G2). Elements (c0, c1) = c0 + c1*u.
+# Affine memory image (halo2curves G2Affine): x.c0 || x.c1 || y.c0 || y.c1,
+# each 32 B LE Montgomery; identity encoded as 128 zero bytes.
+
+def fq2_mul(a, b):
+    a0, a1 = a
+    b0, b1 = b
+    return ((a0 * b0 - a1 * b1) % P, (a0 * b1 + a1 * b0) % P)
+
+def fq2_inv(a):
+    a0, a1 = a
+    d = pow(a0 * a0 + a1 * a1, -1, P)
+    return (a0 * d % P, (-a1) % P * d % P)
+
+B2 = fq2_mul((3, 0), fq2_inv((9, 1)))
+
+# Published BN254 G2 generator (the one halo2curves / EIP-197 use).
+G2_GEN = (
+    (10857046999023057135944570762232829481370756359578518086990519993285655852781,
+     11559732032986387107991004021392285783925812861821192530917403151452391805634),
+    (8495653923123431417604973247489272438418190587263600148770280649306958101930,
+     4082367875863433681332203403145435568316851327593401208105741076214120093531),
+)
+
+def g2_is_on_curve(pt) -> bool:
+    if pt is None:
+        return True
+    x, y = pt
+    rhs = fq2_mul(fq2_mul(x, x), x)
+    rhs = ((rhs[0] + B2[0]) % P, (rhs[1] + B2[1]) % P)
+    lhs = fq2_mul(y, y)
+    return lhs == rhs
+
+assert g2_is_on_curve(G2_GEN), "G2 generator constants wrong"
+
+def g2_neg(pt):
+    if pt is None:
+        return None
+    x, y = pt
+    return (x, ((-y[0]) % P, (-y[1]) % P))
+
+def g2_add(a, b):
+    if a is None:
+        return b
+    if b is None:
+        return a
+    (ax, ay), (bx, by) = a, b
+    if ax == bx:
+        if ((ay[0] + by[0]) % P, (ay[1] + by[1]) % P) == (0, 0):
+            return None
+        num = fq2_mul((3, 0), fq2_mul(ax, ax))
+        lam = fq2_mul(num, fq2_inv(((2 * ay[0]) % P, (2 * ay[1]) % P)))
+    else:
+        num = ((by[0] - ay[0]) % P, (by[1] - ay[1]) % P)
+        den = ((bx[0] - ax[0]) % P, (bx[1] - ax[1]) % P)
+        lam = fq2_mul(num, fq2_inv(den))
+    x3 = fq2_mul(lam, lam)
+    x3 = ((x3[0] - ax[0] - bx[0]) % P, (x3[1] - ax[1] - bx[1]) % P)
+    t = ((ax[0] - x3[0]) % P, (ax[1] - x3[1]) % P)
+    y3 = fq2_mul(lam, t)
+    y3 = ((y3[0] - ay[0]) % P, (y3[1] - ay[1]) % P)
+    return (x3, y3)
+
+def g2_mul(pt, k: int):
+    k %= R
+    acc = None
+    add = pt
+    while k:
+        if k & 1:
+            acc = g2_add(acc, add)
+        add = g2_add(add, add)
+        k >>= 1
+    return acc
+
+def g2_msm(scalars, points):
+    acc = None
+    for s, pt in zip(scalars, points):
+        acc = g2_add(acc, g2_mul(pt, s))
+    return acc
+
+def g2_to_bytes(pt) -> bytes:
+    if pt is None:
+        return b"\x00" * 128
+    x, y = pt
+    return (to_mont_bytes(x[0], P) + to_mont_bytes(x[1], P) +
+            to_mont_bytes(y[0], P) + to_mont_bytes(y[1], P))

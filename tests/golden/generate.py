@@ -26,9 +26,10 @@ import sys
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from bn254_ref import (  # noqa: E402
-    P, R, G1_GEN, fr_root_of_unity, to_mont_bytes, to_canon_bytes,
+    P, R, G1_GEN, G2_GEN, fr_root_of_unity, to_mont_bytes, to_canon_bytes,
     g1_to_bytes, g1_add, g1_mul, g1_neg, g1_is_on_curve, msm, ntt, intt,
-    distribute_powers,
+    distribute_powers, g2_to_bytes, g2_add, g2_mul, g2_neg, g2_msm,
+    g2_is_on_curve,
 )
 
 HERE = os.path.dirname(os.path.abspath(__file__))
@@ -203,11 +204,50 @@ def gen_ntt():
     return out
 
 
+# ------------------------------------------------------------ G2 vectors
+def gen_g2():
+    """G2 (Fq2 twist) — SURVEY §8a minor row: the reference's G2 MSMs are
+    the size <= 2 SRS/verifier ones; the oracle covers the algebra and these
+    fixtures pin it to this independent bigint restatement."""
+    rng = SplitMix64(0x62AF)
+    G = G2_GEN
+    assert g2_is_on_curve(G)
+    pts = [None, G, g2_add(G, G), g2_mul(G, rng.next_fr()), g2_mul(G, rng.next_fr())]
+    cases = []
+    for a in pts:
+        for b in pts:
+            cases.append({"a": hx(g2_to_bytes(a)), "b": hx(g2_to_bytes(b)),
+                          "add": hx(g2_to_bytes(g2_add(a, b)))})
+        cases.append({"a": hx(g2_to_bytes(a)), "b": hx(g2_to_bytes(g2_neg(a))),
+                      "add": hx(g2_to_bytes(None))})
+    muls = []
+    for k in [0, 1, 2, 3, R - 1, R - 2, (1 << 253) - 1, rng.next_fr()]:
+        muls.append({"p": hx(g2_to_bytes(G)), "k": hx(to_canon_bytes(k % R)),
+                     "mul": hx(g2_to_bytes(g2_mul(G, k)))})
+    msms = []
+    Q = g2_mul(G, 0xC0FFEE)
+    for scalars, points in [
+        ([], []),
+        ([rng.next_fr()], [G]),
+        ([rng.next_fr(), rng.next_fr()], [G, Q]),      # the SRS shape (n=2)
+        ([0, rng.next_fr()], [G, Q]),
+        ([R - 1, 1], [G, G]),                          # cancellation
+    ]:
+        msms.append({
+            "n": len(scalars),
+            "scalars_canon": [hx(to_canon_bytes(s)) for s in scalars],
+            "bases": [hx(g2_to_bytes(p)) for p in points],
+            "result": hx(g2_to_bytes(g2_msm(scalars, points))),
+        })
+    return {"add_cases": cases, "mul_cases": muls, "msm_cases": msms}
+
+
 def main():
     fixtures = {
         "fr_arith.json": gen_field(R, "fr"),
         "fq_arith.json": gen_field(P, "fq"),
         "g1.json": gen_g1(),
+        "g2.json": gen_g2(),
         "msm.json": gen_msm(),
         "ntt.json": gen_ntt(),
     }
