@@ -146,6 +146,38 @@ int spectre_gpu_ntt_fr_device(spectre_gpu_ctx*, int dev, void* d_data,
  *   op 0: out = a + b        op 1: out = a - b       op 2: out = a * b
  *   op 3: out = a * c        op 4: out = a + c * b   (c = one Fr element)
  */
+/* ---- gate-expression evaluator (quotient phase) -------------------------
+ * Evaluates one custom-gate expression over every row of a (usually
+ * extended-domain) column set in a single launch — the device-resident
+ * iFFT -> gate-eval -> coset-FFT quotient pipeline needs O(1) launches per
+ * gate instead of per-op round trips.
+ *
+ * program = nops * 3 uint32 words {op, a, b}, a stack machine:
+ *   COL   a=column index, b=(int32) rotation: push cols[a][(row + b*rot_scale) mod n]
+ *   CONST a=constant index: push constants[a]  (challenges, coefficients)
+ *   ADD/SUB/MUL: pop two (second-from-top OP top), push result
+ *   NEG: negate top
+ * Stack depth is validated <= SPECTRE_GATE_MAX_DEPTH. All values are
+ * Montgomery Fr (32 B LE). On exit, for each row:
+ *   y == NULL : out[row]  = result               (first gate)
+ *   y != NULL : out[row] = out[row]*y + result   (halo2's h = h*y + gate)
+ * d_cols is a HOST array of ncols DEVICE pointers; constants/program/y are
+ * host pointers. */
+#define SPECTRE_GATE_OP_COL 0
+#define SPECTRE_GATE_OP_CONST 1
+#define SPECTRE_GATE_OP_ADD 2
+#define SPECTRE_GATE_OP_SUB 3
+#define SPECTRE_GATE_OP_MUL 4
+#define SPECTRE_GATE_OP_NEG 5
+#define SPECTRE_GATE_MAX_DEPTH 8
+int spectre_gpu_fr_gate_eval(spectre_gpu_ctx*, int dev,
+                             const void* const* d_cols, uint32_t ncols,
+                             const uint8_t* constants, uint32_t nconst,
+                             const uint32_t* program, uint32_t nops,
+                             uint64_t n, uint32_t rot_scale,
+                             const uint8_t* y /* 32 B or NULL */,
+                             void* d_out);
+
 #define SPECTRE_VEC_ADD 0
 #define SPECTRE_VEC_SUB 1
 #define SPECTRE_VEC_MUL 2

@@ -515,6 +515,26 @@ int spectre_gpu_ntt_fr(spectre_gpu_ctx* ctx, uint8_t* data, uint32_t log_n,
     return rc;
 }
 
+int spectre_gpu_fr_gate_eval(spectre_gpu_ctx* ctx, int dev,
+                             const void* const* d_cols, uint32_t ncols,
+                             const uint8_t* constants, uint32_t nconst,
+                             const uint32_t* program, uint32_t nops,
+                             uint64_t n, uint32_t rot_scale, const uint8_t* y,
+                             void* d_out) {
+    if (check_dev(ctx, dev)) return -1;
+    if (!d_cols || !program || !nops || !d_out || (nconst && !constants)) {
+        set_err("gate_eval: bad arguments");
+        return -1;
+    }
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    fp256 yv;
+    if (y) ff_from_bytes(yv, y);
+    return fr_gate_eval_device(ctx, dev, (const fp256* const*)d_cols, ncols,
+                               (const fp256*)constants, nconst, program, nops,
+                               n, rot_scale, y ? &yv : nullptr,
+                               (fp256*)d_out);
+}
+
 int spectre_gpu_fr_vec_op(spectre_gpu_ctx* ctx, int dev, int op,
                           const void* d_a, const void* d_b, const uint8_t* c,
                           void* d_out, uint64_t n) {

@@ -97,6 +97,9 @@ struct DeviceState {
     fp256* d_cosetA = nullptr;  // coset base tables (n2 resp. n1 entries)
     fp256* d_cosetB = nullptr;
     size_t coset_cap = 0;
+    // ---- gate-eval scratch (program, constants, column-pointer array) ----
+    uint8_t* d_gate = nullptr;
+    size_t gate_cap = 0;  // bytes
     std::map<std::array<uint8_t, 40>, NttPlan> plans;  // omega||log_n||inverse
 };
 
@@ -140,6 +143,14 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
 
 // msm.hip — sync a slot's stream and deliver pending window sums.
 int msm_slot_drain(spectre_gpu_ctx* ctx, int dev, int slot);
+
+// ntt.hip — gate-expression evaluator over device column buffers
+// (synchronizes). cols = host array of ncols device pointers.
+int fr_gate_eval_device(spectre_gpu_ctx* ctx, int dev,
+                        const fp256* const* cols, uint32_t ncols,
+                        const fp256* consts, uint32_t nconst,
+                        const uint32_t* program, uint32_t nops, uint64_t n,
+                        uint32_t rot_scale, const fp256* y, fp256* d_out);
 
 // ntt.hip — pointwise Fr vector op on device buffers (synchronizes).
 int fr_vec_op_device(spectre_gpu_ctx* ctx, int dev, int op, const fp256* d_a,

@@ -247,6 +247,163 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_axis(
     }
 }
 
+// ---- gate-expression evaluator (quotient phase) -------------
+// One launch evaluates a whole custom-gate expression over every row: a
+// stack machine whose top-of-stack lives in registers and whose lower slots
+// live in LDS (a register-indexed array would spill to scratch — LDS slots
+// are strided [slot][tid] so access is conflict-free). The program is tiny
+// and wave-uniform; columns are read with rotation (row + rot*rot_scale)
+// mod n (n is a power of two).
+#define GATE_THREADS 256
+__global__ __launch_bounds__(GATE_THREADS) void k_fr_gate_eval(
+    const fp256* const* __restrict__ cols, const fp256* __restrict__ consts,
+    const uint32_t* __restrict__ prog, uint32_t nops, uint64_t n,
+    uint32_t rot_scale, int use_y, fp256 y, fp256* __restrict__ out) {
+    extern __shared__ uint4 lds4[];
+    const uint64_t row = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (row >= n) return;
+    const uint32_t T = blockDim.x;
+    const uint64_t mask = n - 1;
+    fp256 tos;
+    ff_set_zero(tos);
+    int depth = 0;
+    auto lds_slot_st = [&](int s, const fp256& v) {
+        uint4 lo, hi;
+        memcpy(&lo, &v.l[0], 16);
+        memcpy(&hi, &v.l[4], 16);
+        lds4[(2 * s) * T + threadIdx.x] = lo;
+        lds4[(2 * s + 1) * T + threadIdx.x] = hi;
+    };
+    auto lds_slot_ld = [&](int s, fp256& v) {
+        uint4 lo = lds4[(2 * s) * T + threadIdx.x];
+        uint4 hi = lds4[(2 * s + 1) * T + threadIdx.x];
+        memcpy(&v.l[0], &lo, 16);
+        memcpy(&v.l[4], &hi, 16);
+    };
+    for (uint32_t pc = 0; pc < nops; pc++) {
+        const uint32_t op = prog[3 * pc];
+        const uint32_t a = prog[3 * pc + 1];
+        const int32_t b = (int32_t)prog[3 * pc + 2];
+        if (op == SPECTRE_GATE_OP_COL || op == SPECTRE_GATE_OP_CONST) {
+            if (depth >= 1) lds_slot_st(depth - 1, tos);
+            if (op == SPECTRE_GATE_OP_COL) {
+                const uint64_t idx =
+                    (row + (uint64_t)((int64_t)b * rot_scale + (int64_t)n)) &
+                    mask;
+                tos = cols[a][idx];
+            } else {
+                tos = consts[a];
+            }
+            depth++;
+        } else if (op == SPECTRE_GATE_OP_NEG) {
+            ff_neg<Fr>(tos, tos);
+        } else {  // binary: (second) op (top)
+            fp256 lhs;
+            lds_slot_ld(depth - 2, lhs);
+            if (op == SPECTRE_GATE_OP_ADD) ff_add<Fr>(tos, lhs, tos);
+            else if (op == SPECTRE_GATE_OP_SUB) ff_sub<Fr>(tos, lhs, tos);
+            else ff_mul<Fr>(tos, lhs, tos);
+            depth--;
+        }
+    }
+    if (use_y) {
+        fp256 acc = out[row];
+        ff_mul<Fr>(acc, acc, y);
+        ff_add<Fr>(tos, acc, tos);
+    }
+    out[row] = tos;
+}
+
+int fr_gate_eval_device(spectre_gpu_ctx* ctx, int dev,
+                        const fp256* const* cols, uint32_t ncols,
+                        const fp256* consts, uint32_t nconst,
+                        const uint32_t* program, uint32_t nops, uint64_t n,
+                        uint32_t rot_scale, const fp256* y, fp256* d_out) {
+    DeviceState& ds = ctx->devs[dev];
+    HIP_TRY(hipSetDevice(ds.device_id));
+    if (n == 0 || (n & (n - 1)) != 0) {
+        set_err("gate_eval: n must be a nonzero power of two");
+        return -3;
+    }
+    // validate the program and compute its maximum stack depth
+    int depth = 0, maxd = 0;
+    for (uint32_t pc = 0; pc < nops; pc++) {
+        const uint32_t op = program[3 * pc];
+        const uint32_t a = program[3 * pc + 1];
+        const int32_t b = (int32_t)program[3 * pc + 2];
+        switch (op) {
+            case SPECTRE_GATE_OP_COL:
+                if (a >= ncols) { set_err("gate_eval: col %u >= %u", a, ncols); return -3; }
+                if ((uint64_t)((int64_t)b * rot_scale < 0
+                                   ? -(int64_t)b * rot_scale
+                                   : (int64_t)b * rot_scale) >= n) {
+                    set_err("gate_eval: rotation %d * %u out of range", b, rot_scale);
+                    return -3;
+                }
+                depth++;
+                break;
+            case SPECTRE_GATE_OP_CONST:
+                if (a >= nconst) { set_err("gate_eval: const %u >= %u", a, nconst); return -3; }
+                depth++;
+                break;
+            case SPECTRE_GATE_OP_NEG:
+                if (depth < 1) { set_err("gate_eval: NEG on empty stack"); return -3; }
+                break;
+            case SPECTRE_GATE_OP_ADD:
+            case SPECTRE_GATE_OP_SUB:
+            case SPECTRE_GATE_OP_MUL:
+                if (depth < 2) { set_err("gate_eval: binary op underflow at pc %u", pc); return -3; }
+                depth--;
+                break;
+            default:
+                set_err("gate_eval: bad opcode %u at pc %u", op, pc);
+                return -3;
+        }
+        if (depth > maxd) maxd = depth;
+        if (maxd > SPECTRE_GATE_MAX_DEPTH) {
+            set_err("gate_eval: stack depth %d > %d", maxd, SPECTRE_GATE_MAX_DEPTH);
+            return -3;
+        }
+    }
+    if (depth != 1) {
+        set_err("gate_eval: program leaves %d values on the stack (need 1)", depth);
+        return -3;
+    }
+    // device staging: [col ptrs][constants][program], one cached buffer
+    const size_t ptr_b = (size_t)ncols * sizeof(fp256*);
+    const size_t con_b = (size_t)nconst * sizeof(fp256);
+    const size_t prg_b = (size_t)nops * 12;
+    const size_t need = ptr_b + con_b + prg_b;
+    if (ds.gate_cap < need) {
+        if (ds.d_gate) { (void)hipFree(ds.d_gate); ds.d_gate = nullptr; }
+        ds.gate_cap = 0;
+        HIP_TRY(hipMalloc(&ds.d_gate, need));
+        ds.gate_cap = need;
+    }
+    HIP_TRY(hipMemcpyAsync(ds.d_gate, cols, ptr_b, hipMemcpyHostToDevice,
+                           ds.stream));
+    if (con_b)
+        HIP_TRY(hipMemcpyAsync(ds.d_gate + ptr_b, consts, con_b,
+                               hipMemcpyHostToDevice, ds.stream));
+    HIP_TRY(hipMemcpyAsync(ds.d_gate + ptr_b + con_b, program, prg_b,
+                           hipMemcpyHostToDevice, ds.stream));
+    fp256 yv;
+    ff_set_zero(yv);
+    if (y) yv = *y;
+    const uint32_t lds_bytes =
+        (maxd > 1 ? (uint32_t)(maxd - 1) * GATE_THREADS * 32u : 0u);
+    hipLaunchKernelGGL(k_fr_gate_eval,
+                       dim3((uint32_t)((n + GATE_THREADS - 1) / GATE_THREADS)),
+                       dim3(GATE_THREADS), lds_bytes, ds.stream,
+                       (const fp256* const*)ds.d_gate,
+                       (const fp256*)(ds.d_gate + ptr_b),
+                       (const uint32_t*)(ds.d_gate + ptr_b + con_b), nops, n,
+                       rot_scale, y ? 1 : 0, yv, d_out);
+    HIP_TRY(hipStreamSynchronize(ds.stream));
+    HIP_TRY(hipGetLastError());
+    return 0;
+}
+
 // ---- pointwise Fr vector ops (quotient-phase gate-eval glue) --------------
 __global__ void k_fr_vec_op(int op, const fp256* __restrict__ a,
                             const fp256* __restrict__ b, fp256 c,

@@ -80,6 +80,12 @@ def load_library() -> ctypes.CDLL:
     ]
     lib.spectre_gpu_msm_slot_wait.restype = c.c_int
     lib.spectre_gpu_msm_slot_wait.argtypes = [c.c_void_p, c.c_int, c.c_int]
+    lib.spectre_gpu_fr_gate_eval.restype = c.c_int
+    lib.spectre_gpu_fr_gate_eval.argtypes = [
+        c.c_void_p, c.c_int, c.POINTER(c.c_void_p), c.c_uint32, c.c_void_p,
+        c.c_uint32, c.POINTER(c.c_uint32), c.c_uint32, c.c_uint64, c.c_uint32,
+        c.c_void_p, c.c_void_p,
+    ]
     lib.spectre_gpu_msm_g1_combine.restype = c.c_int
     lib.spectre_gpu_msm_g1_combine.argtypes = [c.c_void_p, c.c_uint32, c.c_void_p]
     lib.spectre_gpu_ntt_fr.restype = c.c_int
@@ -283,6 +289,30 @@ class SpectreGpu:
             self._ctx, dev, ctypes.c_void_p(d_data), log_n, om,
             1 if inverse else 0, cg)
         self._check(rc, "ntt_fr_device")
+
+    # gate-expression evaluator opcodes (spectre_gpu.h)
+    GATE_COL, GATE_CONST, GATE_ADD, GATE_SUB, GATE_MUL, GATE_NEG = range(6)
+
+    def gate_eval(self, d_cols: list[int], constants: bytes,
+                  program: list[tuple[int, int, int]], n: int,
+                  rot_scale: int = 1, y: bytes | None = None,
+                  d_out: int = 0, dev: int = 0) -> None:
+        """Evaluate a gate expression over n rows into device buffer d_out
+        (out = out*y + v when y is given). program = [(op, a, rot)]."""
+        cols = (ctypes.c_void_p * len(d_cols))(*d_cols)
+        prog = (ctypes.c_uint32 * (3 * len(program)))()
+        for i, (op, a, b) in enumerate(program):
+            prog[3 * i] = op
+            prog[3 * i + 1] = a
+            prog[3 * i + 2] = b & 0xFFFFFFFF
+        con = ((ctypes.c_uint8 * len(constants)).from_buffer_copy(constants)
+               if constants else None)
+        yb = (ctypes.c_uint8 * 32).from_buffer_copy(y) if y else None
+        rc = self._lib.spectre_gpu_fr_gate_eval(
+            self._ctx, dev, cols, len(d_cols), con,
+            len(constants) // 32, prog, len(program), n, rot_scale, yb,
+            ctypes.c_void_p(d_out))
+        self._check(rc, "fr_gate_eval")
 
     VEC_ADD, VEC_SUB, VEC_MUL, VEC_SCALE, VEC_ADD_SCALED = range(5)
 

@@ -403,6 +403,119 @@ def test_ntt_2pow25_roundtrip(gpu, oracle, golden):
     gpu.free(d)
 
 
+def _ref_gate_eval(cols_int, consts_int, program, n, rot_scale, r_mod,
+                   y_int=None, prev=None):
+    """Bigint restatement of the gate-expression evaluator semantics
+    (the checker for SURVEY §8f-3): stack machine per row, rotations
+    (row + rot*rot_scale) mod n, out = prev*y + v when y given."""
+    out = []
+    for row in range(n):
+        st = []
+        for op, a, b in program:
+            if op == 0:
+                st.append(cols_int[a][(row + b * rot_scale) % n])
+            elif op == 1:
+                st.append(consts_int[a])
+            elif op == 5:
+                st[-1] = (-st[-1]) % r_mod
+            else:
+                rhs = st.pop()
+                lhs = st.pop()
+                st.append((lhs + rhs) % r_mod if op == 2 else
+                          (lhs - rhs) % r_mod if op == 3 else
+                          lhs * rhs % r_mod)
+        v = st[0]
+        if y_int is not None:
+            v = (prev[row] * y_int + v) % r_mod
+        out.append(v)
+    return out
+
+
+def test_gate_eval_vs_reference(gpu, oracle):
+    """Gate-expression evaluator (quotient phase): the
+    halo2-base flex-gate shape q*(a + b*c - d) over rotations of one
+    column, plus random programs, vs the bigint restatement."""
+    import random
+    R = 21888242871839275222246405745257275088548364400416034343698204186575808495617
+    n = 1 << 10
+    rng = random.Random(8015)
+
+    def fr_vec(ints):
+        return b"".join(
+            oracle.fr_from_canonical(v.to_bytes(32, "little")) for v in ints)
+
+    def to_ints(raw):
+        return [int.from_bytes(
+            oracle.fr_to_canonical(raw[32 * i:32 * (i + 1)]), "little")
+            for i in range(len(raw) // 32)]
+
+    ncols = 3
+    cols_int = [[rng.randrange(R) for _ in range(n)] for _ in range(ncols)]
+    d_cols = []
+    for ci in cols_int:
+        d = gpu.malloc(32 * n)
+        gpu.upload(d, fr_vec(ci))
+        d_cols.append(d)
+    d_out = gpu.malloc(32 * n)
+
+    G = gpu
+    # flex gate: q * (a + b*c - d) with a,b,c,d = rotations 0..3 of col 1,
+    # q = col 0 (rot_scale 4 as the extended domain would use)
+    flex = [(G.GATE_COL, 0, 0),
+            (G.GATE_COL, 1, 0), (G.GATE_COL, 1, 1), (G.GATE_COL, 1, 2),
+            (G.GATE_MUL, 0, 0), (G.GATE_ADD, 0, 0),
+            (G.GATE_COL, 1, 3), (G.GATE_SUB, 0, 0),
+            (G.GATE_MUL, 0, 0)]
+    G.gate_eval(d_cols, b"", flex, n, rot_scale=4, d_out=d_out)
+    want = _ref_gate_eval(cols_int, [], flex, n, 4, R)
+    assert to_ints(bytes(G.download(d_out, 32 * n))) == want
+
+    # accumulate form out = out*y + v, with constants and negation
+    y = rng.randrange(R)
+    consts = [rng.randrange(R) for _ in range(2)]
+    prog2 = [(G.GATE_CONST, 0, 0), (G.GATE_COL, 2, -1), (G.GATE_MUL, 0, 0),
+             (G.GATE_CONST, 1, 0), (G.GATE_NEG, 0, 0), (G.GATE_ADD, 0, 0)]
+    G.gate_eval(d_cols, fr_vec(consts), prog2, n, rot_scale=1,
+                y=fr_vec([y]), d_out=d_out)
+    want2 = _ref_gate_eval(cols_int, consts, prog2, n, 1, R, y, want)
+    assert to_ints(bytes(G.download(d_out, 32 * n))) == want2
+
+    # random well-formed programs to max depth
+    for trial in range(6):
+        prog, depth, maxd = [], 0, 0
+        consts = [rng.randrange(R) for _ in range(3)]
+        while len(prog) < 20 or depth != 1:
+            if depth >= 2 and (depth >= 7 or rng.random() < 0.5):
+                prog.append((rng.choice([G.GATE_ADD, G.GATE_SUB, G.GATE_MUL]),
+                             0, 0))
+                depth -= 1
+            elif depth >= 1 and rng.random() < 0.1:
+                prog.append((G.GATE_NEG, 0, 0))
+            elif rng.random() < 0.7:
+                prog.append((G.GATE_COL, rng.randrange(ncols),
+                             rng.randrange(-3, 4)))
+                depth += 1
+            else:
+                prog.append((G.GATE_CONST, rng.randrange(3), 0))
+                depth += 1
+            maxd = max(maxd, depth)
+        G.gate_eval(d_cols, fr_vec(consts), prog, n, rot_scale=2, d_out=d_out)
+        want = _ref_gate_eval(cols_int, consts, prog, n, 2, R)
+        assert to_ints(bytes(G.download(d_out, 32 * n))) == want, trial
+
+    # validation: malformed programs are rejected loudly
+    import pytest as _pytest
+    for bad in ([(G.GATE_ADD, 0, 0)],                     # underflow
+                [(G.GATE_COL, 99, 0)],                    # bad column
+                [(G.GATE_COL, 0, 0), (G.GATE_COL, 0, 0)],  # 2 left on stack
+                [(G.GATE_COL, 0, 0)] * 9):                # depth > max
+        with _pytest.raises(RuntimeError):
+            G.gate_eval(d_cols, b"", bad, n, d_out=d_out)
+
+    for d in d_cols + [d_out]:
+        gpu.free(d)
+
+
 def test_fr_vec_ops_vs_oracle(gpu, oracle):
     """Pointwise Fr vector ops (quotient gate-eval glue) vs the oracle."""
     n = 4097
