@@ -33,8 +33,11 @@ R = 2188824287183927522224640574525727508854836440041603434369820418657580849561
 # structure for the pinned configs (CALLCOUNTS.md; committee24 = estimate).
 # Stage order follows the prover: advice commits, lookup permuted commits,
 # permutation z + lookup product commits, random poly, per-column iFFT +
-# coset-FFT for the quotient, extended icoset + h commits, SHPLONK.
-# Consecutive same-basis commits batch (the fused batch MSM API).
+# coset-FFT for the quotient, the extended-domain constraint evaluation
+# (modeled as (1 + 3L + NZ) gate-eval passes of the flex-gate program with
+# y-accumulation — one fused custom-gate pass + per-lookup and per-z-chunk
+# constraint passes), extended icoset + h commits, SHPLONK. Consecutive
+# same-basis commits batch (the fused batch MSM API).
 TRACES = {
     # sync-step k=20: 19 advice cols (15 base + 2 lookup-advice + 2 spread),
     # L=3 lookups, NZ=11 z-chunks, j=3 h pieces, ext domain 2^22.
@@ -46,6 +49,7 @@ TRACES = {
         ("vanishing random",   "msm",   20,  1),
         ("quotient iFFT",      "intt",  20, 40),
         ("quotient coset-FFT", "coset", 22, 40),
+        ("quotient gate-eval", "gate",  22, 21),
         ("h extended icoset",  "icoset", 22, 1),
         ("quotient h commits", "msm",   20,  3),
         ("SHPLONK multiopen",  "msm",   20,  2),
@@ -59,6 +63,7 @@ TRACES = {
         ("vanishing random",   "msm",   23,  1),
         ("quotient iFFT",      "intt",  23,  8),
         ("quotient coset-FFT", "coset", 25,  8),
+        ("quotient gate-eval", "gate",  25,  6),
         ("h extended icoset",  "icoset", 25, 1),
         ("quotient h commits", "msm",   23,  3),
         ("SHPLONK multiopen",  "msm",   23,  2),
@@ -74,6 +79,7 @@ TRACES = {
         ("vanishing random",   "msm",   24,  1),
         ("quotient iFFT",      "intt",  24,  8),
         ("quotient coset-FFT", "coset", 26,  8),
+        ("quotient gate-eval", "gate",  26,  6),
         ("h extended icoset",  "icoset", 26, 1),
         ("quotient h commits", "msm",   24,  3),
         ("SHPLONK multiopen",  "msm",   24,  2),
@@ -139,6 +145,13 @@ def main():
     gpu.upload(d_v20, vec20)
     d_v22 = gpu.malloc(32 << EXT_K)
     gpu.upload(d_v22, vec22)
+    # gate-eval column set on the extended domain (3 columns suffice for the
+    # flex-gate shape; contents are timing-irrelevant)
+    d_gcols = [d_v22, gpu.malloc(32 << EXT_K), gpu.malloc(32 << EXT_K)]
+    d_gout = gpu.malloc(32 << EXT_K)
+    for d in d_gcols[1:]:
+        gpu.upload(d, vec22)
+    y_acc = oracle.gen_fr_vector(1, 99)
     w20, w22 = omega_for(K), omega_for(EXT_K)
     w20i, w22i = oracle.fr_inv(w20), oracle.fr_inv(w22)
 
@@ -151,6 +164,16 @@ def main():
                 gpu.msm_batch_device(d_b, d_batch, count, n)
             else:
                 gpu.msm_shard_device(d_b, d_s20 if log_n == K else d_s22, n)
+        elif kind == "gate":
+            G = gpu
+            flex = [(G.GATE_COL, 0, 0),
+                    (G.GATE_COL, 1, 0), (G.GATE_COL, 1, 1),
+                    (G.GATE_COL, 1, 2), (G.GATE_MUL, 0, 0),
+                    (G.GATE_ADD, 0, 0), (G.GATE_COL, 2, 3),
+                    (G.GATE_SUB, 0, 0), (G.GATE_MUL, 0, 0)]
+            for i in range(count):
+                G.gate_eval(d_gcols, b"", flex, 1 << log_n, rot_scale=4,
+                            y=None if i == 0 else y_acc, d_out=d_gout)
         else:
             d_v = d_v20 if log_n == K else d_v22
             for _ in range(count):
