@@ -115,15 +115,17 @@ def main():
         """Run k complete MSMs; returns the last result."""
         result = None
         if pipelined:
-            prev = None
+            depth = max(2, int(os.environ.get("SPECTRE_PIPE_SLOTS", "2")))
+            pend = []
             for _ in range(k):
-                buf, slot = gpu.msm_shard_device_async(d_b, d_s, m)
-                if prev is not None:
-                    gpu.msm_slot_wait(prev[1])
-                    result = ffi.combine_partials(bytes(prev[0]), 1)
-                prev = (buf, slot)
-            gpu.msm_slot_wait(prev[1])
-            result = ffi.combine_partials(bytes(prev[0]), 1)
+                pend.append(gpu.msm_shard_device_async(d_b, d_s, m))
+                if len(pend) >= depth:
+                    buf, slot = pend.pop(0)
+                    gpu.msm_slot_wait(slot)
+                    result = ffi.combine_partials(bytes(buf), 1)
+            for buf, slot in pend:
+                gpu.msm_slot_wait(slot)
+                result = ffi.combine_partials(bytes(buf), 1)
         else:
             for _ in range(k):
                 result = step()

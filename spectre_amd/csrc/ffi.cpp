@@ -9,6 +9,7 @@
 #include "internal.hpp"
 #include "spectre_gpu.h"
 #include <cstdarg>
+#include <cstdlib>
 #include <cstdio>
 
 thread_local std::string g_last_error;
@@ -250,8 +251,13 @@ int spectre_gpu_msm_g1_shard_device_async(spectre_gpu_ctx* ctx, int dev,
     if (check_dev(ctx, dev)) return -1;
     std::lock_guard<std::recursive_mutex> lk(ctx->mu);
     DeviceState& ds = ctx->devs[dev];
+    static const int kSlots = []() {
+        const char* e = getenv("SPECTRE_PIPE_SLOTS");
+        int v = e ? atoi(e) : 2;
+        return v < 1 ? 1 : (v > 3 ? 3 : v);
+    }();
     const int slot = ds.next_slot;
-    ds.next_slot ^= 1;
+    ds.next_slot = (ds.next_slot + 1) % kSlots;
     *out_slot = slot;
     return msm_batch_shard_device(ctx, dev, (const g1_affine*)d_bases,
                                   (const uint8_t*)d_scalars, 1, n, flags,
@@ -261,7 +267,7 @@ int spectre_gpu_msm_g1_shard_device_async(spectre_gpu_ctx* ctx, int dev,
 
 int spectre_gpu_msm_slot_wait(spectre_gpu_ctx* ctx, int dev, int slot) {
     if (check_dev(ctx, dev)) return -1;
-    if (slot < 0 || slot > 1) {
+    if (slot < 0 || slot > 2) {
         set_err("slot_wait: slot %d out of range", slot);
         return -1;
     }

@@ -24,8 +24,12 @@
 #define MSM_NB_TOTAL (MSM_NWIN * MSM_BPW)
 #define MSM_SKIP_KEY MSM_NB_TOTAL          // sorts after all real keys
 #define MSM_SORT_BITS 20                   // key range < 2^20
+#ifndef MSM_CHUNK
 #define MSM_CHUNK 8                        // buckets per reduction thread
+#endif
+#ifndef MSM_ACC_E
 #define MSM_ACC_E 64                       // sorted entries per acc thread
+#endif
 
 struct NttPlan {
     fp256* tw1 = nullptr;  // butterfly twiddles, axis 1
@@ -75,7 +79,7 @@ struct MsmSlot {
 struct DeviceState {
     int device_id = 0;
     hipStream_t stream = nullptr;
-    MsmSlot slots[2];
+    MsmSlot slots[3];  // >=2 enables the async pipeline; 3rd for depth-3 A/B
     int next_slot = 0;  // round-robin for the async API
     uint8_t* d_scalars = nullptr;
     size_t scal_cap = 0;  // bytes

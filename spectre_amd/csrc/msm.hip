@@ -322,7 +322,7 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            int slot) {
     DeviceState& dstate = ctx->devs[dev];
     HIP_TRY(hipSetDevice(dstate.device_id));
-    if (slot < 0 || slot > 1) {
+    if (slot < 0 || slot > 2) {
         set_err("msm: slot %d out of range", slot);
         return -3;
     }
@@ -448,7 +448,7 @@ int msm_shard_device(spectre_gpu_ctx* ctx, int dev, const g1_affine* d_bases,
 // caller memory). Idempotent when nothing is pending.
 int msm_slot_drain(spectre_gpu_ctx* ctx, int dev, int slot) {
     DeviceState& dstate = ctx->devs[dev];
-    if (slot < 0 || slot > 1) {
+    if (slot < 0 || slot > 2) {
         set_err("slot_drain: slot %d out of range", slot);
         return -1;
     }

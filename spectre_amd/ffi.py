@@ -8,7 +8,9 @@ import ctypes
 import os
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
-_LIB = os.path.join(_HERE, "libspectre_gpu.so")
+# SPECTRE_GPU_LIB overrides the library path (tuning-variant A/B builds)
+_LIB = os.environ.get("SPECTRE_GPU_LIB",
+                      os.path.join(_HERE, "libspectre_gpu.so"))
 
 SCALARS_MONTGOMERY = 0
 SCALARS_CANONICAL = 1
