@@ -251,9 +251,12 @@ int spectre_gpu_msm_g1_shard_device_async(spectre_gpu_ctx* ctx, int dev,
     if (check_dev(ctx, dev)) return -1;
     std::lock_guard<std::recursive_mutex> lk(ctx->mu);
     DeviceState& ds = ctx->devs[dev];
+    // measured on MI355X (r2): depth 3 = 325.6 MSM(2^20)/s vs depth 2 =
+    // 300 vs unpipelined = 240 — three in flight keep the machine fed
+    // through each call's host-combine + sync gap.
     static const int kSlots = []() {
         const char* e = getenv("SPECTRE_PIPE_SLOTS");
-        int v = e ? atoi(e) : 2;
+        int v = e ? atoi(e) : 3;
         return v < 1 ? 1 : (v > 3 ? 3 : v);
     }();
     const int slot = ds.next_slot;
