@@ -85,25 +85,79 @@ __global__ void k_pow_table(fp256 base, fp256* __restrict__ out,
 
 // in-LDS DIF butterflies over L = 2^logL elements; twL[j] = (root)^j, j < L/2.
 // On exit lds[s] holds DFT output index bitrev(s, logL).
+//
+// Levels run in FUSED PAIRS (radix-4 rounds): a thread loads the quad
+// (i0, i0+h/2, i0+h, i0+3h/2), applies level h then level h/2 in registers,
+// and writes back once — HALF the LDS round trips and HALF the barriers of
+// per-level radix-2, with the exact same field operations (bit-identical
+// results; finite-field ops are exact). An odd level count runs one plain
+// radix-2 level first. Build with -DSPECTRE_NTT_RADIX2 to restore the
+// per-level schedule (A/B variant).
 __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
                         uint32_t logL) {
     const uint32_t L = 1u << logL;
-    for (uint32_t h = L >> 1; h >= 1; h >>= 1) {
-        const uint32_t stride = (L >> 1) / h;
-        for (uint32_t p = threadIdx.x; p < (L >> 1); p += blockDim.x) {
-            uint32_t blk = p / h, j = p % h;
-            uint32_t i0 = blk * 2 * h + j, i1 = i0 + h;
-            fp256 u, v, t, w;
-            lds_ld(lds4, H, i0, u);
-            lds_ld(lds4, H, i1, v);
-            ff_add<Fr>(w, u, v);
-            lds_st(lds4, H, i0, w);
-            ff_sub<Fr>(t, u, v);
-            ff_mul<Fr>(t, t, twL[(uint64_t)j * stride]);
-            lds_st(lds4, H, i1, t);
+    uint32_t h = L >> 1;
+#ifndef SPECTRE_NTT_RADIX2
+    if (logL & 1)  // one radix-2 level so the rest pairs evenly
+#endif
+    {
+        for (; h >= 1; h >>= 1) {
+            const uint32_t stride = (L >> 1) / h;
+            for (uint32_t p = threadIdx.x; p < (L >> 1); p += blockDim.x) {
+                uint32_t blk = p / h, j = p % h;
+                uint32_t i0 = blk * 2 * h + j, i1 = i0 + h;
+                fp256 u, v, t, w;
+                lds_ld(lds4, H, i0, u);
+                lds_ld(lds4, H, i1, v);
+                ff_add<Fr>(w, u, v);
+                lds_st(lds4, H, i0, w);
+                ff_sub<Fr>(t, u, v);
+                ff_mul<Fr>(t, t, twL[(uint64_t)j * stride]);
+                lds_st(lds4, H, i1, t);
+            }
+            __syncthreads();
+#ifndef SPECTRE_NTT_RADIX2
+            h >>= 1;
+            break;
+#endif
+        }
+    }
+#ifndef SPECTRE_NTT_RADIX2
+    for (; h >= 2; h >>= 2) {
+        const uint32_t hh = h >> 1;               // second level of the pair
+        const uint32_t s_h = (L >> 1) / h;        // level-h twiddle stride
+        for (uint32_t p = threadIdx.x; p < (L >> 2); p += blockDim.x) {
+            const uint32_t blk = p / hh, j = p % hh;
+            const uint32_t i0 = blk * 2 * h + j;
+            fp256 a, b, c, d, t;
+            lds_ld(lds4, H, i0, a);
+            lds_ld(lds4, H, i0 + hh, b);
+            lds_ld(lds4, H, i0 + h, c);
+            lds_ld(lds4, H, i0 + h + hh, d);
+            // level h: pairs (a,c) at offset j and (b,d) at offset j+hh
+            fp256 a1, b1, c1, d1;
+            ff_add<Fr>(a1, a, c);
+            ff_sub<Fr>(t, a, c);
+            ff_mul<Fr>(c1, t, twL[(uint64_t)j * s_h]);
+            ff_add<Fr>(b1, b, d);
+            ff_sub<Fr>(t, b, d);
+            ff_mul<Fr>(d1, t, twL[(uint64_t)(j + hh) * s_h]);
+            // level h/2: pairs (a1,b1) and (c1,d1), both at offset j,
+            // twiddle stride 2*s_h
+            ff_add<Fr>(a, a1, b1);
+            ff_sub<Fr>(t, a1, b1);
+            ff_mul<Fr>(b, t, twL[(uint64_t)j * 2 * s_h]);
+            ff_add<Fr>(c, c1, d1);
+            ff_sub<Fr>(t, c1, d1);
+            ff_mul<Fr>(d, t, twL[(uint64_t)j * 2 * s_h]);
+            lds_st(lds4, H, i0, a);
+            lds_st(lds4, H, i0 + hh, b);
+            lds_st(lds4, H, i0 + h, c);
+            lds_st(lds4, H, i0 + h + hh, d);
         }
         __syncthreads();
     }
+#endif
 }
 
 // pass A: column DFTs. grid.x = n2; LDS = n1 elements.
