@@ -98,7 +98,11 @@ __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
     const uint32_t L = 1u << logL;
     uint32_t h = L >> 1;
 #ifndef SPECTRE_NTT_RADIX2
-    if (logL & 1)  // one radix-2 level so the rest pairs evenly
+    // radix-4 rounds only when every thread gets a quad — with fewer quads
+    // than threads the idle 3/4 cost more than the saved round trips
+    // (measured: 2^20 0.40 -> 0.49 ms regression on 1024-elem tiles).
+    const bool r4 = (L >> 2) >= blockDim.x;
+    if (!r4 || (logL & 1))  // radix-2 for odd level count (one level) or all
 #endif
     {
         for (; h >= 1; h >>= 1) {
@@ -117,8 +121,7 @@ __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
             }
             __syncthreads();
 #ifndef SPECTRE_NTT_RADIX2
-            h >>= 1;
-            break;
+            if (r4) { h >>= 1; break; }
 #endif
         }
     }
