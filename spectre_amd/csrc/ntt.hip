@@ -40,6 +40,11 @@
 #define THREADS 256       // pow-table builder
 #define NTT_THREADS 1024  // NTT passes: 16 waves/block so one LDS-resident
                           // block still puts 4 waves on every SIMD
+#ifdef SPECTRE_NTT_RADIX2
+static constexpr bool kForceRadix2 = true;   // A/B variant build
+#else
+static constexpr bool kForceRadix2 = false;
+#endif
 #define TW_LOW_BITS 12
 #define TW_LOW_MASK 0xfffu
 
@@ -93,17 +98,16 @@ __global__ void k_pow_table(fp256 base, fp256* __restrict__ out,
 // results; finite-field ops are exact). An odd level count runs one plain
 // radix-2 level first. Build with -DSPECTRE_NTT_RADIX2 to restore the
 // per-level schedule (A/B variant).
+// R4 is a compile-time variant: the radix-4 quad path costs VGPRs/code even
+// when branched over at run time (measured 2^20 0.40 -> 0.46 ms with a
+// run-time flag), so the host launches the <true> instantiation only when
+// the tile gives every thread a quad (L >= 4*threads).
+template <bool R4>
 __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
                         uint32_t logL) {
     const uint32_t L = 1u << logL;
     uint32_t h = L >> 1;
-#ifndef SPECTRE_NTT_RADIX2
-    // radix-4 rounds only when every thread gets a quad — with fewer quads
-    // than threads the idle 3/4 cost more than the saved round trips
-    // (measured: 2^20 0.40 -> 0.49 ms regression on 1024-elem tiles).
-    const bool r4 = (L >> 2) >= blockDim.x;
-    if (!r4 || (logL & 1))  // radix-2 for odd level count (one level) or all
-#endif
+    if (!R4 || (logL & 1))  // radix-2: all levels, or one to make pairs even
     {
         for (; h >= 1; h >>= 1) {
             const uint32_t stride = (L >> 1) / h;
@@ -120,12 +124,10 @@ __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
                 lds_st(lds4, H, i1, t);
             }
             __syncthreads();
-#ifndef SPECTRE_NTT_RADIX2
-            if (r4) { h >>= 1; break; }
-#endif
+            if constexpr (R4) { h >>= 1; break; }
         }
     }
-#ifndef SPECTRE_NTT_RADIX2
+    if constexpr (!R4) return;
     for (; h >= 2; h >>= 2) {
         const uint32_t hh = h >> 1;               // second level of the pair
         const uint32_t s_h = (L >> 1) / h;        // level-h twiddle stride
@@ -160,10 +162,10 @@ __device__ void lds_dif(uint4* lds4, uint32_t H, const fp256* __restrict__ twL,
         }
         __syncthreads();
     }
-#endif
 }
 
 // pass A: column DFTs. grid.x = n2; LDS = n1 elements.
+template <bool R4>
 __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
                           const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw1,
@@ -187,7 +189,7 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
         lds_st(lds4, H, s, v);
     }
     __syncthreads();
-    lds_dif(lds4, H, tw1, log_n1);
+    lds_dif<R4>(lds4, H, tw1, log_n1);
     // store with inter-pass twiddle omega^(c * t1), t1 = bitrev(s)
     for (uint32_t s = threadIdx.x; s < n1; s += blockDim.x) {
         uint32_t t1 = bitrev(s, log_n1);
@@ -202,6 +204,7 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_col(
 
 // pass B: row DFTs + transposed store. grid.x = n1; LDS = n2 elements.
 // in == out is safe only when n1 == 1 (single workgroup).
+template <bool R4>
 __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
                           const fp256* __restrict__ in, fp256* __restrict__ out,
                           const fp256* __restrict__ tw2,
@@ -224,7 +227,7 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
         lds_st(lds4, H, s, v);
     }
     __syncthreads();
-    lds_dif(lds4, H, tw2, log_n2);
+    lds_dif<R4>(lds4, H, tw2, log_n2);
     for (uint32_t s = threadIdx.x; s < n2; s += blockDim.x) {
         uint32_t t2 = bitrev(s, log_n2);
         fp256 v;
@@ -249,6 +252,7 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_row(
 // w^(t*m); pass2 (axis b, root w^(AC)) by w^(A*tB*c) = w^((t<<logA)*(m&(C-1)));
 // pass3 (axis c, root w^(AB)) scatters to out[tA + A*tB + AB*t] and carries
 // the n^{-1} scale / inverse-coset factor. All exponents < n <= 2^28 (u32).
+template <bool R4>
 __global__ __launch_bounds__(NTT_THREADS) void k_ntt_axis(
     const fp256* __restrict__ in, fp256* __restrict__ out,
     const fp256* __restrict__ twL, const fp256* __restrict__ T1,
@@ -275,7 +279,7 @@ __global__ __launch_bounds__(NTT_THREADS) void k_ntt_axis(
         lds_st(lds4, H, s, v);
     }
     __syncthreads();
-    lds_dif(lds4, H, twL, logL);
+    lds_dif<R4>(lds4, H, twL, logL);
     for (uint32_t s = threadIdx.x; s < L; s += blockDim.x) {
         const uint32_t t = bitrev(s, logL);
         fp256 v, f;
@@ -634,15 +638,20 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         auto tdim = [](uint32_t L) {
             return L < NTT_THREADS ? (L < 64 ? 64u : L) : (uint32_t)NTT_THREADS;
         };
-        hipLaunchKernelGGL(k_ntt_axis, dim3(B * C), dim3(tdim(A)),
+        // radix-4 instantiation only when every thread gets a quad
+        auto axis = [](uint32_t L, uint32_t threads) {
+            return (!kForceRadix2 && L >= 4 * threads) ? k_ntt_axis<true>
+                                                       : k_ntt_axis<false>;
+        };
+        hipLaunchKernelGGL(axis(A, tdim(A)), dim3(B * C), dim3(tdim(A)),
                            LDS_BYTES(A), st, d_data, ds.d_ntt_tmp, plan->tw1,
                            T1, T2, fwd_cT1, fwd_cT2, /*coset_on_load=*/1, one,
                            0, kA, kB + kC, /*post_mode=*/1, 0, 0, 0, 0, 0);
-        hipLaunchKernelGGL(k_ntt_axis, dim3(A * C), dim3(tdim(B)),
+        hipLaunchKernelGGL(axis(B, tdim(B)), dim3(A * C), dim3(tdim(B)),
                            LDS_BYTES(B), st, ds.d_ntt_tmp, ds.d_ntt_tmp,
                            plan->tw2, T1, T2, nullptr, nullptr, 0, one, 0, kB,
                            kC, /*post_mode=*/2, kA, kC, 0, 0, 0);
-        hipLaunchKernelGGL(k_ntt_axis, dim3(A * B), dim3(tdim(C)),
+        hipLaunchKernelGGL(axis(C, tdim(C)), dim3(A * B), dim3(tdim(C)),
                            LDS_BYTES(C), st, ds.d_ntt_tmp, d_data, plan->tw3,
                            T1, T2, inv_cT1, inv_cT2, /*coset_on_load=*/0,
                            scale, inverse ? 1 : 0, kC, 0, /*post_mode=*/0, 0,
@@ -650,12 +659,16 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
     } else if (plan->k1 > 0) {
         const uint32_t tc = n1 < NTT_THREADS ? (n1 < 64 ? 64 : n1) : NTT_THREADS;
         const uint32_t tr = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
-        hipLaunchKernelGGL(k_ntt_col, dim3(n2), dim3(tc),
+        hipLaunchKernelGGL((!kForceRadix2 && n1 >= 4 * tc) ? k_ntt_col<true>
+                                                           : k_ntt_col<false>,
+                           dim3(n2), dim3(tc),
                            LDS_BYTES(n1), st, d_data,
                            ds.d_ntt_tmp, plan->tw1, plan->twB,
                            plan->twB + t1n, fwd_cT1, fwd_cT2, plan->k1,
                            plan->k2);
-        hipLaunchKernelGGL(k_ntt_row, dim3(n1), dim3(tr),
+        hipLaunchKernelGGL((!kForceRadix2 && n2 >= 4 * tr) ? k_ntt_row<true>
+                                                           : k_ntt_row<false>,
+                           dim3(n1), dim3(tr),
                            LDS_BYTES(n2), st, ds.d_ntt_tmp,
                            d_data, plan->tw2, inv_cT1, inv_cT2,
                            /*coset_on_load=*/0, scale, inverse ? 1 : 0,
@@ -665,7 +678,9 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         const fp256* cc1 = coset_gen ? cT1 : nullptr;
         const fp256* cc2 = coset_gen ? cT2 : nullptr;
         const uint32_t t1p = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
-        hipLaunchKernelGGL(k_ntt_row, dim3(1), dim3(t1p),
+        hipLaunchKernelGGL((!kForceRadix2 && n2 >= 4 * t1p) ? k_ntt_row<true>
+                                                            : k_ntt_row<false>,
+                           dim3(1), dim3(t1p),
                            LDS_BYTES(n2), st, d_data, d_data,
                            plan->tw2, cc1, cc2,
                            /*coset_on_load=*/inverse ? 0 : 1, scale,
