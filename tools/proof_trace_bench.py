@@ -68,6 +68,18 @@ TRACES = {
         ("quotient h commits", "msm",   23,  3),
         ("SHPLONK multiopen",  "msm",   23,  2),
     ]),
+    # keygen (create_pk) for sync-step k=20 — ESTIMATE:
+    # commit_lagrange + iFFT per fixed polynomial (~15 flex-gate q columns +
+    # 1 constants + 2 spread tables; halo2-base materializes gate selectors
+    # as fixed columns) and per permutation sigma (P=21), plus pk cosets for
+    # each and l0/l_last/l_active_row. Confirmed by the capture shim like
+    # the proof counts (CALLCOUNTS.md); same kernels, keygen-shaped call mix.
+    "keygen20": (20, 22, [
+        ("fixed commits",      "msm",   20, 18),
+        ("sigma commits",      "msm",   20, 21),
+        ("fixed+sigma iFFT",   "intt",  20, 39),
+        ("pk cosets",          "coset", 22, 42),
+    ]),
     # committee-update aggregation (K=24): same formulas at 2^24/2^26.
     # (The committee-update k=20 leaf circuit itself is NOT modeled: its
     # zkevm-hashes SHA column count is unresolvable offline, CALLCOUNTS.md.)
