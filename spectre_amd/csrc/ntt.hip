@@ -553,8 +553,14 @@ static int get_plan(DeviceState& ds, const fp256& omega, uint32_t log_n,
         build_tables(ds, w3, p.tw3, 1u << (p.k3 - 1));
     } else {
         // balanced 2-pass split; k2 >= k1 so the contiguous row pass gets
-        // the bigger tile
-        p.k2 = log_n <= 12 ? log_n : (log_n + 1) / 2;
+        // the bigger tile. SPECTRE_NTT_SPLIT=max forces k2 = 12 (A/B: the
+        // 4096-elem row tile runs radix-4 rounds, at the cost of a smaller
+        // column tile).
+        const char* sp = getenv("SPECTRE_NTT_SPLIT");
+        if (sp && sp[0] == 'm' && log_n > 12)
+            p.k2 = 12;
+        else
+            p.k2 = log_n <= 12 ? log_n : (log_n + 1) / 2;
         p.k1 = log_n - p.k2;
         const uint32_t n1 = 1u << p.k1, n2 = 1u << p.k2;
         fp256 w1, w2;
