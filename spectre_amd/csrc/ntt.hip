@@ -552,12 +552,16 @@ static int get_plan(DeviceState& ds, const fp256& omega, uint32_t log_n,
         build_tables(ds, w2, p.tw2, 1u << (p.k2 - 1));
         build_tables(ds, w3, p.tw3, 1u << (p.k3 - 1));
     } else {
-        // balanced 2-pass split; k2 >= k1 so the contiguous row pass gets
-        // the bigger tile. SPECTRE_NTT_SPLIT=max forces k2 = 12 (A/B: the
-        // 4096-elem row tile runs radix-4 rounds, at the cost of a smaller
-        // column tile).
+        // 2-pass split, k2 >= k1 (the contiguous row pass gets the bigger
+        // tile). Measured on MI355X: k2 = 12 (4096-elem row tiles -> radix-4
+        // rounds) wins for log_n <= 20 (2^20 fwd 0.40 -> 0.32 ms) but LOSES
+        // at 2^22 ((10,12) 1.40 vs balanced (11,11) 1.31 — the column pass's
+        // longer stride outweighs the row gain), so: k2=12 up to log_n 20,
+        // balanced above (log_n 23/24 balance to k2=12 anyway).
+        // SPECTRE_NTT_SPLIT=max / =bal force either choice for A/B.
         const char* sp = getenv("SPECTRE_NTT_SPLIT");
-        if (sp && sp[0] == 'm' && log_n > 12)
+        const bool maxk2 = sp ? (sp[0] == 'm') : (log_n <= 20);
+        if (maxk2 && log_n > 12)
             p.k2 = 12;
         else
             p.k2 = log_n <= 12 ? log_n : (log_n + 1) / 2;
