@@ -151,6 +151,103 @@ FF_HD void g1j_add_ip(g1_jac& p, const g1_jac& q) {
     p.X = H;
 }
 
+// ---- XYZZ coordinates (kept as a MEASURED NEGATIVE for bucket accumulation)
+// x = X/ZZ, y = Y/ZZZ with ZZ^3 == ZZZ^2. Mixed add (mADD-2008-s) is
+// 8M + 2S and ~4 add/sub vs Jacobian madd-2007-bl's 7M + 4S and ~7, BUT
+// using it as the k_bucket_acc accumulator (with per-run g1xyzz_to_jac
+// conversion) measured SLOWER on MI355X: 152 VGPRs / 3 waves/SIMD vs 84 /
+// 4-5 for the Jacobian form — bucket_acc 2.31 -> 3.09 ms at n=2^20. The
+// occupancy loss beats the ~9% multiply saving. Retained (bit-exact,
+// unused on the hot path) for lower-pressure contexts. Identity: ZZ == 0.
+struct g1_xyzz {
+    fp256 X, Y, ZZ, ZZZ;
+};
+
+FF_HD void g1x_set_inf(g1_xyzz& p) {
+    ff_set_one<Fq>(p.X);
+    ff_set_one<Fq>(p.Y);
+    ff_set_zero(p.ZZ);
+    ff_set_zero(p.ZZZ);
+}
+FF_HD bool g1x_is_inf(const g1_xyzz& p) { return ff_is_zero(p.ZZ); }
+FF_HD void g1x_from_affine(g1_xyzz& o, const g1_affine& p) {
+    if (g1a_is_inf(p)) { g1x_set_inf(o); return; }
+    o.X = p.x;
+    o.Y = p.y;
+    ff_set_one<Fq>(o.ZZ);
+    ff_set_one<Fq>(o.ZZZ);
+}
+
+// p = 2p, dbl-2008-s-1 (a = 0), in place
+FF_HD void g1x_dbl_ip(g1_xyzz& p) {
+    if (g1x_is_inf(p)) return;
+    fp256 U, V, W, S, M, t;
+    ff_add<Fq>(U, p.Y, p.Y);   // U = 2Y
+    ff_sqr<Fq>(V, U);          // V = U^2
+    ff_mul<Fq>(W, U, V);       // W = U*V
+    ff_mul<Fq>(S, p.X, V);     // S = X*V
+    ff_sqr<Fq>(M, p.X);
+    ff_add<Fq>(t, M, M);
+    ff_add<Fq>(M, t, M);       // M = 3X^2
+    ff_sqr<Fq>(t, M);
+    ff_sub<Fq>(t, t, S);
+    ff_sub<Fq>(t, t, S);       // X' = M^2 - 2S
+    ff_sub<Fq>(S, S, t);
+    ff_mul<Fq>(S, M, S);       // M*(S - X')
+    ff_mul<Fq>(p.Y, W, p.Y);   // W*Y
+    ff_sub<Fq>(p.Y, S, p.Y);   // Y' = M*(S-X') - W*Y
+    p.X = t;
+    ff_mul<Fq>(p.ZZ, p.ZZ, V);
+    ff_mul<Fq>(p.ZZZ, p.ZZZ, W);
+}
+
+// p += q (q affine), mADD-2008-s, in place
+FF_HD void g1x_madd_ip(g1_xyzz& p, const g1_affine& q) {
+    if (g1a_is_inf(q)) return;
+    if (g1x_is_inf(p)) { g1x_from_affine(p, q); return; }
+    fp256 U2, S2, P, R, PP, PPP, Q, t;
+    ff_mul<Fq>(U2, q.x, p.ZZ);
+    ff_mul<Fq>(S2, q.y, p.ZZZ);
+    ff_sub<Fq>(P, U2, p.X);
+    ff_sub<Fq>(R, S2, p.Y);
+    if (ff_is_zero(P)) {
+        if (ff_is_zero(R)) { g1x_dbl_ip(p); return; }
+        g1x_set_inf(p);
+        return;
+    }
+    ff_sqr<Fq>(PP, P);
+    ff_mul<Fq>(PPP, P, PP);
+    ff_mul<Fq>(Q, p.X, PP);
+    ff_sqr<Fq>(t, R);
+    ff_sub<Fq>(t, t, PPP);
+    ff_sub<Fq>(t, t, Q);
+    ff_sub<Fq>(t, t, Q);       // X3 = R^2 - PPP - 2Q
+    ff_sub<Fq>(Q, Q, t);       // Q - X3
+    ff_mul<Fq>(Q, R, Q);
+    ff_mul<Fq>(p.Y, p.Y, PPP);
+    ff_sub<Fq>(p.Y, Q, p.Y);   // Y3 = R*(Q - X3) - Y1*PPP
+    p.X = t;
+    ff_mul<Fq>(p.ZZ, p.ZZ, PP);
+    ff_mul<Fq>(p.ZZZ, p.ZZZ, PPP);
+}
+
+// xyzz -> Jacobian: Z = ZZZ/ZZ would need an inversion; instead pick
+// Z = ZZ*ZZZ, then X_j = x*Z^2 = X*ZZ*ZZZ^2 and Y_j = y*Z^3 = Y*ZZ^2*ZZZ^2
+// ... wait: x = X/ZZ, X_j = x*Z^2 = X*ZZ*ZZZ^2; y = Y/ZZZ,
+// Y_j = y*Z^3 = Y*ZZ^3*ZZZ^2. 5 muls + squares, amortized once per run.
+FF_HD void g1xyzz_to_jac(g1_jac& o, const g1_xyzz& p) {
+    if (g1x_is_inf(p)) { g1j_set_inf(o); return; }
+    fp256 zz2, zzz2, t;
+    ff_sqr<Fq>(zzz2, p.ZZZ);              // ZZZ^2
+    ff_mul<Fq>(o.Z, p.ZZ, p.ZZZ);         // Z = ZZ*ZZZ
+    ff_mul<Fq>(t, p.ZZ, zzz2);            // ZZ*ZZZ^2
+    ff_mul<Fq>(o.X, p.X, t);              // X_j
+    ff_sqr<Fq>(zz2, p.ZZ);                // ZZ^2
+    ff_mul<Fq>(t, zz2, p.ZZ);             // ZZ^3
+    ff_mul<Fq>(t, t, zzz2);               // ZZ^3*ZZZ^2
+    ff_mul<Fq>(o.Y, p.Y, t);              // Y_j
+}
+
 // ---- copy-style wrappers (host/ffi convenience) ----
 FF_HD void g1j_dbl(g1_jac& o, const g1_jac& p) {
     o = p;

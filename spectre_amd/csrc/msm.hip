@@ -128,6 +128,11 @@ PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
     uint64_t hi = lo + MSM_ACC_E;
     if (hi > ent) hi = ent;
     uint32_t k = keys[lo];
+    // MEASURED NEGATIVE RESULT (r2): accumulating runs in XYZZ
+    // (mADD-2008-s, 8M+2S — see g1.hpp) with a per-run Jacobian conversion
+    // is bit-exact but SLOWER: 152 VGPRs / 3 waves/SIMD vs the Jacobian
+    // accumulator's 84 / 4-5, and bucket_acc went 2.31 -> 3.09 ms at
+    // n=2^20 — the occupancy loss outweighs the ~9% fewer multiplies.
     g1_jac acc;
     g1j_set_inf(acc);
     bool first = true;
