@@ -232,9 +232,9 @@ FF_HD void g1x_madd_ip(g1_xyzz& p, const g1_affine& q) {
 }
 
 // xyzz -> Jacobian: Z = ZZZ/ZZ would need an inversion; instead pick
-// Z = ZZ*ZZZ, then X_j = x*Z^2 = X*ZZ*ZZZ^2 and Y_j = y*Z^3 = Y*ZZ^2*ZZZ^2
-// ... wait: x = X/ZZ, X_j = x*Z^2 = X*ZZ*ZZZ^2; y = Y/ZZZ,
-// Y_j = y*Z^3 = Y*ZZ^3*ZZZ^2. 5 muls + squares, amortized once per run.
+// Z = ZZ*ZZZ, so X_j = x*Z^2 = (X/ZZ)*(ZZ*ZZZ)^2 = X*ZZ*ZZZ^2 and
+// Y_j = y*Z^3 = (Y/ZZZ)*(ZZ*ZZZ)^3 = Y*ZZ^3*ZZZ^2. Inversion-free,
+// amortized once per run.
 FF_HD void g1xyzz_to_jac(g1_jac& o, const g1_xyzz& p) {
     if (g1x_is_inf(p)) { g1j_set_inf(o); return; }
     fp256 zz2, zzz2, t;
