@@ -332,6 +332,12 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
         return -3;
     }
     MsmSlot& ds = dstate.slots[slot];
+    if (!sync && ds.pending_dst) {
+        // a second in-flight call on one slot would overwrite the pending
+        // D2H silently — fail loudly instead (callers must slot_wait first)
+        set_err("msm: slot %d already has a call in flight", slot);
+        return -3;
+    }
     if (!ds.stream) {
         if (slot == 0) ds.stream = dstate.stream;
         else HIP_TRY(hipStreamCreate(&ds.stream));
