@@ -537,6 +537,64 @@ def test_gate_eval_vs_reference(gpu, oracle):
         gpu.free(d)
 
 
+def test_quotient_pipeline_device_resident(gpu, oracle):
+    """The whole quotient phase device-resident: per-column
+    iFFT -> coset-FFT, gate expression over the coset values, inverse-coset
+    iFFT of the result — one upload, one download, bit-exact against the
+    oracle NTTs + a bigint evaluation of the same expression."""
+    R = 21888242871839275222246405745257275088548364400416034343698204186575808495617
+    log_n = 10
+    n = 1 << log_n
+    # omega from the module ROOT28 (verified vs fixtures in test_root_consistency)
+    w = ROOT28
+    for _ in range(28 - log_n):
+        w = oracle.fr_mul(w, w)
+    wi = oracle.fr_inv(w)
+    g5 = oracle.fr_from_canonical((5).to_bytes(32, "little"))
+    g5i = oracle.fr_inv(g5)
+
+    cols = [oracle.gen_fr_vector(n, 7000 + i) for i in range(3)]
+    d_cols = []
+    for cvec in cols:
+        d = gpu.malloc(32 * n)
+        gpu.upload(d, cvec)
+        d_cols.append(d)
+    d_out = gpu.malloc(32 * n)
+
+    # device: evaluations -> coeffs -> coset evals, per column
+    for d in d_cols:
+        gpu.ntt_device(d, log_n, wi, inverse=True)
+        gpu.ntt_device(d, log_n, w, coset_gen=g5)
+    G = gpu
+    flex = [(G.GATE_COL, 0, 0),
+            (G.GATE_COL, 1, 0), (G.GATE_COL, 1, 1), (G.GATE_COL, 2, 2),
+            (G.GATE_MUL, 0, 0), (G.GATE_ADD, 0, 0),
+            (G.GATE_COL, 2, 3), (G.GATE_SUB, 0, 0),
+            (G.GATE_MUL, 0, 0)]
+    G.gate_eval(d_cols, b"", flex, n, rot_scale=4, d_out=d_out)
+    gpu.ntt_device(d_out, log_n, wi, inverse=True, coset_gen=g5i)
+    got = bytes(gpu.download(d_out, 32 * n))
+
+    # oracle: identical sequence
+    ref_cosets = []
+    for cvec in cols:
+        coeff = oracle.ntt(cvec, log_n, wi, inverse=True)
+        ref_cosets.append(oracle.ntt(coeff, log_n, w, coset_gen=g5))
+
+    def to_ints(raw):
+        return [int.from_bytes(
+            oracle.fr_to_canonical(raw[32 * i:32 * (i + 1)]), "little")
+            for i in range(n)]
+    ci = [to_ints(c) for c in ref_cosets]
+    ev = _ref_gate_eval(ci, [], flex, n, 4, R)
+    ev_bytes = b"".join(
+        oracle.fr_from_canonical(v.to_bytes(32, "little")) for v in ev)
+    want = oracle.ntt(ev_bytes, log_n, wi, inverse=True, coset_gen=g5i)
+    assert got == want
+    for d in d_cols + [d_out]:
+        gpu.free(d)
+
+
 def test_fr_vec_ops_vs_oracle(gpu, oracle):
     """Pointwise Fr vector ops (quotient gate-eval glue) vs the oracle."""
     n = 4097
