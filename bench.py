@@ -218,7 +218,8 @@ def main():
                 "n": N,
                 "seed": SEED,
                 "parallelism": f"shard{n_gpus}" + (f"+{backend}" if world > 1 else ""),
-                "pipeline_depth": 2 if pipelined else 1,
+                "pipeline_depth": (max(2, int(os.environ.get(
+                    "SPECTRE_PIPE_SLOTS", "3"))) if pipelined else 1),
                 "scalars": "canonical",
             },
             "roofline": roofline,
