@@ -536,11 +536,21 @@ static int get_plan(DeviceState& ds, const fp256& omega, uint32_t log_n,
     build_tables(ds, omega, p.twB, t1n);
     build_tables(ds, wT2, p.twB + t1n, t2n);
     if (log_n > 24 || (force3 && log_n >= 3)) {
-        // three-pass split n = A*B*C, each axis <= 2^12
-        const uint32_t q = log_n / 3, r = log_n % 3;
-        p.k1 = q + (r > 0);  // kA
-        p.k2 = q + (r > 1);  // kB
-        p.k3 = q;            // kC
+        // three-pass split n = A*B*C, each axis <= 2^12 (balanced), or
+        // SPECTRE_NTT_SPLIT3=max: kA = 12 so pass A gets 4096-elem tiles
+        // (radix-4 rounds) and B/C split the rest.
+        const char* s3 = getenv("SPECTRE_NTT_SPLIT3");
+        if (s3 && s3[0] == 'm' && log_n > 14) {
+            p.k1 = 12;
+            const uint32_t rest = log_n - 12;
+            p.k2 = (rest + 1) / 2;
+            p.k3 = rest - p.k2;
+        } else {
+            const uint32_t q = log_n / 3, r = log_n % 3;
+            p.k1 = q + (r > 0);  // kA
+            p.k2 = q + (r > 1);  // kB
+            p.k3 = q;            // kC
+        }
         fp256 w1, w2, w3;
         host_pow_u32(w1, omega, 1u << (p.k2 + p.k3));  // w^(B*C)
         host_pow_u32(w2, omega, 1u << (p.k1 + p.k3));  // w^(A*C)

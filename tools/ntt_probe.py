@@ -25,7 +25,9 @@ def omega_for(log_n):
 def main():
     gpu = SpectreGpu([0])
     g = oracle.fr_from_canonical((5).to_bytes(32, "little"))
-    for log_n in [20, 22, 23, 24]:
+    sizes = ([int(a) for a in sys.argv[1:]] if len(sys.argv) > 1
+             else [20, 22, 23, 24])
+    for log_n in sizes:
         n = 1 << log_n
         data = oracle.gen_fr_vector(min(n, 1 << 20), 7)
         data = data * (n // (1 << min(log_n, 20)))
