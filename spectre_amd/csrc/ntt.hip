@@ -539,8 +539,11 @@ static int get_plan(DeviceState& ds, const fp256& omega, uint32_t log_n,
         // three-pass split n = A*B*C, each axis <= 2^12 (balanced), or
         // SPECTRE_NTT_SPLIT3=max: kA = 12 so pass A gets 4096-elem tiles
         // (radix-4 rounds) and B/C split the rest.
+        // measured: kA=12 (radix-4 axis-A tiles) is -6% at 2^26, neutral at
+        // 2^25 -> default for log_n >= 26; SPECTRE_NTT_SPLIT3=max/bal force.
         const char* s3 = getenv("SPECTRE_NTT_SPLIT3");
-        if (s3 && s3[0] == 'm' && log_n > 14) {
+        const bool max3 = s3 ? (s3[0] == 'm') : (log_n >= 26);
+        if (max3 && log_n > 14) {
             p.k1 = 12;
             const uint32_t rest = log_n - 12;
             p.k2 = (rest + 1) / 2;
