@@ -44,11 +44,28 @@ fi
 cp "$HERE/spectre.rs" "$SRC/spectre.rs"
 printf '\npub mod spectre;\n' >> "$SRC/lib.rs"
 
-# 4. cargo features (+ link script for the gpu feature)
+# 4. cargo features (+ link script for the gpu feature).
+# halo2_proofs is a TRANSITIVE dependency of Spectre (via halo2-base), and
+# cargo cannot enable a transitive dep's features from the CLI — so the
+# shim features are toggled as DEFAULTS of the patched (local) crate:
+#   apply_patch.sh CRATE            -> default = capture only (CPU-safe;
+#                                      capture is inert without
+#                                      SPECTRE_CAPTURE set at run time)
+#   apply_patch.sh CRATE gpu        -> default = capture + gpu dispatch
+MODE="${2:-capture}"
+ADD='"spectre-capture"'
+[ "$MODE" = gpu ] && ADD='"spectre-capture", "spectre-gpu"'
 if grep -q '^\[features\]' "$CRATE/Cargo.toml"; then
-    sed -i '/^\[features\]/a spectre-gpu = []\nspectre-capture = []' "$CRATE/Cargo.toml"
+    sed -i "/^\[features\]/a spectre-gpu = []\nspectre-capture = []" "$CRATE/Cargo.toml"
 else
     printf '\n[features]\nspectre-gpu = []\nspectre-capture = []\n' >> "$CRATE/Cargo.toml"
+fi
+# MERGE into any existing default list (upstream defaults must survive);
+# TOML tolerates the trailing comma when the list was empty.
+if grep -q '^default *= *\[' "$CRATE/Cargo.toml"; then
+    sed -i "s/^default *= *\[/default = [$ADD, /" "$CRATE/Cargo.toml"
+else
+    sed -i "/^\[features\]/a default = [$ADD]" "$CRATE/Cargo.toml"
 fi
 if [ ! -f "$CRATE/build.rs" ]; then
     cat > "$CRATE/build.rs" <<'EOF'
@@ -63,4 +80,4 @@ fn main() {
 }
 EOF
 fi
-echo "applied. Build with --features spectre-capture and/or spectre-gpu."
+echo "applied (defaults include spectre-capture$([ "$MODE" = gpu ] && echo /spectre-gpu)); SPECTRE_CAPTURE env gates logging at run time."

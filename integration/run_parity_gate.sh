@@ -30,7 +30,7 @@ SPECTRE="${1:?usage: run_parity_gate.sh /path/to/spectre /path/to/halo2_proofs [
 HALO2="${2:?need halo2_proofs path}"
 REPO="${3:-$(cd "$(dirname "$0")/.." && pwd)}"
 
-"$(dirname "$0")/apply_patch.sh" "$HALO2"
+"$(dirname "$0")/apply_patch.sh" "$HALO2"   # capture-only defaults (leg 1/2)
 
 grep -q 'halo2_proofs.*path' "$SPECTRE/Cargo.toml" || cat >> "$SPECTRE/Cargo.toml" <<EOF
 
@@ -43,7 +43,7 @@ EOF
 CAP="$REPO/gpurun_out/capture_step20"
 rm -rf "$CAP" && mkdir -p "$CAP"
 ( cd "$SPECTRE" && SPECTRE_CAPTURE="$CAP" cargo test -r -p lightclient-circuits \
-    --features halo2_proofs/spectre-capture test_step_proofgen -- --nocapture )
+    test_step_proofgen -- --nocapture )
 echo "captured $(wc -l < "$CAP/calls.jsonl") seam calls"
 
 # leg 2: reference-pinned golden fixtures + oracle check
@@ -54,11 +54,13 @@ python3 "$REPO/tools/capture_to_golden.py" "$CAP" --out "$REPO/tests/golden/capt
 if command -v rocminfo >/dev/null 2>&1; then
     CAPG="$REPO/gpurun_out/capture_step20_gpu"
     rm -rf "$CAPG" && mkdir -p "$CAPG"
+    # flip the patched crate's defaults to include the GPU dispatch
+    grep -q '"spectre-gpu"' <(grep '^default' "$HALO2/Cargo.toml") || \
+        sed -i 's/^default *= *\[/default = ["spectre-gpu", /' "$HALO2/Cargo.toml"
     ( cd "$SPECTRE" && SPECTRE_CAPTURE="$CAPG" \
         SPECTRE_GPU_LIB_DIR="$REPO/spectre_amd" \
         LD_LIBRARY_PATH="$REPO/spectre_amd:${LD_LIBRARY_PATH:-}" \
         cargo test -r -p lightclient-circuits \
-        --features halo2_proofs/spectre-capture,halo2_proofs/spectre-gpu \
         test_step_proofgen -- --nocapture )
     python3 "$REPO/tools/capture_to_golden.py" "$CAPG" --diff "$CAP"
 else
