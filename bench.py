@@ -74,9 +74,25 @@ def main():
     scalars, bases = oracle.gen_msm_inputs(N, SEED, fast=True)
     log(f"input gen {time.time() - t0:.1f}s")
 
-    lo = N * rank // world
-    hi = N * (rank + 1) // world
+    # Sharding strategy (SPECTRE_SHARD=window|chunk):
+    #  * window (default when world divides 16): rank r computes windows
+    #    [r*16/world, ...) over ALL points — bucket work AND the reduction
+    #    tail divide by world, and the exchange is a pure allgather of
+    #    DISJOINT window sums. This is what makes the strong-scaling curve
+    #    scale past the fixed per-rank tail of chunk sharding.
+    #  * chunk: the SURVEY §8e contiguous scalar-chunk split (kept for A/B;
+    #    per-rank tail is fixed so it stalls at high world sizes).
+    shard_mode = os.environ.get("SPECTRE_SHARD", "window")
+    if world > 1 and (ffi.NUM_WINDOWS % world != 0):
+        shard_mode = "chunk"
+    if shard_mode == "window":
+        lo, hi = 0, N            # every rank holds the full inputs
+    else:
+        lo = N * rank // world
+        hi = N * (rank + 1) // world
     m = hi - lo
+    w_cnt = ffi.NUM_WINDOWS // world
+    w_lo = rank * w_cnt
 
     gpu = SpectreGpu([local_rank] if world > 1 else list(range(args.gpus)))
     # upload this rank's shard; resident in HBM before the timed region
@@ -86,17 +102,28 @@ def main():
     gpu.upload(d_s, scalars[32 * lo:32 * hi])
     dev = torch.device(f"cuda:{local_rank}")
 
+    def exchange(partials: bytes):
+        """allgather this rank's partial blob and combine (both modes)."""
+        if dist is None:
+            return (ffi.combine_window_partials(partials, 1)
+                    if shard_mode == "window" else
+                    ffi.combine_partials(partials, 1))
+        t = torch.frombuffer(bytearray(partials), dtype=torch.uint8)
+        if backend == "nccl":
+            t = t.to(dev)  # <=1.5 KiB window sums over RCCL/xGMI
+        gath = [torch.empty_like(t) for _ in range(world)]
+        dist.all_gather(gath, t)
+        blob = b"".join(g.cpu().numpy().tobytes() for g in gath)
+        if shard_mode == "window":
+            return ffi.combine_window_partials(blob, world)
+        return ffi.combine_partials(blob, world)
+
     def step():
-        partials = gpu.msm_shard_device(d_b, d_s, m)
-        if dist is not None:
-            t = torch.frombuffer(bytearray(partials), dtype=torch.uint8)
-            if backend == "nccl":
-                t = t.to(dev)  # 1.5 KiB window sums over RCCL/xGMI
-            gath = [torch.empty_like(t) for _ in range(world)]
-            dist.all_gather(gath, t)
-            blob = b"".join(g.cpu().numpy().tobytes() for g in gath)
-            return ffi.combine_partials(blob, world)
-        return ffi.combine_partials(partials, 1)
+        if shard_mode == "window":
+            partials = gpu.msm_shard_windows_device(d_b, d_s, m, w_lo, w_cnt)
+        else:
+            partials = gpu.msm_shard_device(d_b, d_s, m)
+        return exchange(partials)
 
     # in-library multi-dev fallback (single process, --gpus>1, no torchrun)
     if world == 1 and args.gpus > 1:
@@ -109,23 +136,34 @@ def main():
     # commits arrive exactly like this). Every step still runs the complete
     # MSM incl. the host combine; K steps fully drain inside the timed
     # region. Depth 3 measured best (325 vs 300 vs 240 MSM/s).
-    pipelined = world == 1 and args.gpus == 1 and not args.no_pipeline
+    pipelined = args.gpus == 1 and not args.no_pipeline
 
     def run_steps(k):
         """Run k complete MSMs; returns the last result."""
         result = None
         if pipelined:
+            # under torchrun the allgather of step i-1 overlaps this rank's
+            # compute of step i (depth 2 across the collective)
             depth = max(2, int(os.environ.get("SPECTRE_PIPE_SLOTS", "3")))
+            if dist is not None:
+                depth = 2
+
+            def enqueue():
+                if shard_mode == "window" and dist is not None:
+                    return gpu.msm_shard_windows_device_async(
+                        d_b, d_s, m, w_lo, w_cnt)
+                return gpu.msm_shard_device_async(d_b, d_s, m)
+
             pend = []
             for _ in range(k):
-                pend.append(gpu.msm_shard_device_async(d_b, d_s, m))
+                pend.append(enqueue())
                 if len(pend) >= depth:
                     buf, slot = pend.pop(0)
                     gpu.msm_slot_wait(slot)
-                    result = ffi.combine_partials(bytes(buf), 1)
+                    result = exchange(bytes(buf))
             for buf, slot in pend:
                 gpu.msm_slot_wait(slot)
-                result = ffi.combine_partials(bytes(buf), 1)
+                result = exchange(bytes(buf))
         else:
             for _ in range(k):
                 result = step()

@@ -123,6 +123,32 @@ int spectre_gpu_msm_slot_wait(spectre_gpu_ctx*, int dev, int slot);
 int spectre_gpu_msm_g1_combine(const uint8_t* partials, uint32_t nshards,
                                uint8_t out_affine[64]);
 
+/* ---- window-sharded multi-GPU --------------------------------------------
+ * Scalar-chunk sharding (above) divides only the bucket work: the
+ * per-window reduction tail is fixed per rank, so strong scaling stalls.
+ * Window sharding instead gives rank r of R the windows
+ * [r*NUM_WINDOWS/R, (r+1)*NUM_WINDOWS/R) over ALL n points: bucket work AND
+ * tail divide by R, and the exchange is a pure allgather of DISJOINT window
+ * sums (no reduction at all). R must divide NUM_WINDOWS (1,2,4,8,16).
+ * out_partials = w_cnt * 96 B Jacobian sums. */
+int spectre_gpu_msm_g1_shard_windows_device(spectre_gpu_ctx*, int dev,
+                                            const void* d_bases,
+                                            const void* d_scalars, uint64_t n,
+                                            uint32_t flags, uint32_t w_lo,
+                                            uint32_t w_cnt,
+                                            uint8_t* out_partials);
+/* async variant on the pipeline slots (same contract as
+ * spectre_gpu_msm_g1_shard_device_async). */
+int spectre_gpu_msm_g1_shard_windows_device_async(
+    spectre_gpu_ctx*, int dev, const void* d_bases, const void* d_scalars,
+    uint64_t n, uint32_t flags, uint32_t w_lo, uint32_t w_cnt,
+    uint8_t* out_partials, int* out_slot);
+/* Concatenated rank-ordered slices (nshards * (NUM_WINDOWS/nshards) * 96 B)
+ * -> affine result. nshards must divide NUM_WINDOWS. */
+int spectre_gpu_msm_g1_combine_windows(const uint8_t* partials,
+                                       uint32_t nshards,
+                                       uint8_t out_affine[64]);
+
 /* ---- NTT ----------------------------------------------------------------
  * In-place radix-2 NTT over Fr, Montgomery-form data, exactly halo2's
  * best_fft / EvaluationDomain semantics:

@@ -268,6 +268,55 @@ int spectre_gpu_msm_g1_shard_device_async(spectre_gpu_ctx* ctx, int dev,
                                   /*sync=*/false, slot);
 }
 
+// window-sharded multi-GPU: rank r of R computes windows
+// [r*NWIN/R, (r+1)*NWIN/R) over ALL n points. out_partials = w_cnt * 96 B.
+int spectre_gpu_msm_g1_shard_windows_device(spectre_gpu_ctx* ctx, int dev,
+                                            const void* d_bases,
+                                            const void* d_scalars, uint64_t n,
+                                            uint32_t flags, uint32_t w_lo,
+                                            uint32_t w_cnt,
+                                            uint8_t* out_partials) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    return msm_batch_windows_device(ctx, dev, (const g1_affine*)d_bases,
+                                    (const uint8_t*)d_scalars, 1, n, flags,
+                                    w_lo, w_cnt, (g1_jac*)out_partials);
+}
+
+int spectre_gpu_msm_g1_shard_windows_device_async(
+    spectre_gpu_ctx* ctx, int dev, const void* d_bases, const void* d_scalars,
+    uint64_t n, uint32_t flags, uint32_t w_lo, uint32_t w_cnt,
+    uint8_t* out_partials, int* out_slot) {
+    if (check_dev(ctx, dev)) return -1;
+    std::lock_guard<std::recursive_mutex> lk(ctx->mu);
+    DeviceState& ds = ctx->devs[dev];
+    static const int kSlots = []() {
+        const char* e = getenv("SPECTRE_PIPE_SLOTS");
+        int v = e ? atoi(e) : 3;
+        return v < 1 ? 1 : (v > 3 ? 3 : v);
+    }();
+    const int slot = ds.next_slot;
+    ds.next_slot = (ds.next_slot + 1) % kSlots;
+    *out_slot = slot;
+    return msm_batch_windows_device(ctx, dev, (const g1_affine*)d_bases,
+                                    (const uint8_t*)d_scalars, 1, n, flags,
+                                    w_lo, w_cnt, (g1_jac*)out_partials,
+                                    nullptr, /*sync=*/false, slot);
+}
+
+// assemble disjoint per-shard window slices (shard i = NWIN/nshards
+// consecutive windows, rank order) and finish with the window Horner.
+int spectre_gpu_msm_g1_combine_windows(const uint8_t* partials,
+                                       uint32_t nshards,
+                                       uint8_t out_affine[64]) {
+    if (!partials || nshards == 0 || MSM_NWIN % nshards != 0) {
+        set_err("combine_windows: nshards must divide %d", MSM_NWIN);
+        return -1;
+    }
+    winsums_to_affine((const g1_jac*)partials, out_affine);
+    return 0;
+}
+
 int spectre_gpu_msm_slot_wait(spectre_gpu_ctx* ctx, int dev, int slot) {
     if (check_dev(ctx, dev)) return -1;
     if (slot < 0 || slot > 2) {

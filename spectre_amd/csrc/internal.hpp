@@ -145,6 +145,18 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            g1_jac* winsums_host, double* stage_ms = nullptr,
                            bool sync = true, int slot = 0);
 
+// msm.hip — window-ranged pipeline: only windows [w_lo, w_lo+w_cnt) are
+// decomposed/accumulated (window-sharded multi-GPU: bucket work AND the
+// reduction tail divide by the shard count; the carry recoding still runs
+// over all windows). winsums_host receives nbatch * w_cnt Jacobian sums.
+int msm_batch_windows_device(spectre_gpu_ctx* ctx, int dev,
+                             const g1_affine* d_bases,
+                             const uint8_t* d_scalars, uint32_t nbatch,
+                             uint64_t n, uint32_t flags, uint32_t w_lo,
+                             uint32_t w_cnt, g1_jac* winsums_host,
+                             double* stage_ms = nullptr, bool sync = true,
+                             int slot = 0);
+
 // msm.hip — sync a slot's stream and deliver pending window sums.
 int msm_slot_drain(spectre_gpu_ctx* ctx, int dev, int slot);
 

@@ -46,13 +46,20 @@
 #define PT_KERNEL __global__ __launch_bounds__(THREADS, 2)
 
 // ---- kernel 1: signed window decomposition --------------------------------
+// w_lo/w_cnt: emit only windows [w_lo, w_lo+w_cnt) — window-sharded
+// multi-GPU (each rank owns disjoint windows over ALL points, so bucket
+// work AND the reduction tail divide by the rank count; the exchange is a
+// pure allgather of disjoint window sums). The signed recoding carry runs
+// over ALL windows regardless (it propagates from window 0 upward).
 __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
-                             uint32_t nbatch, int canonical,
+                             uint32_t nbatch, int canonical, uint32_t w_lo,
+                             uint32_t w_cnt,
                              uint32_t* __restrict__ keys,
                              uint32_t* __restrict__ vals) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    const uint32_t skip_key = nbatch * MSM_NB_TOTAL;
+    const uint32_t nb_local = w_cnt * MSM_BPW;
+    const uint32_t skip_key = nbatch * nb_local;
     for (uint32_t b = 0; b < nbatch; b++) {
         fp256 s;
         ff_from_bytes(s, scalars + 32 * (b * n + i));
@@ -74,16 +81,19 @@ __global__ void k_msm_digits(const uint8_t* __restrict__ scalars, uint64_t n,
                 key = skip_key;
             } else if (d <= half) {  // positive digit, magnitude d
                 carry = 0;
-                key = b * MSM_NB_TOTAL + w * MSM_BPW + (d - 1);
+                key = b * nb_local + ((uint32_t)w - w_lo) * MSM_BPW + (d - 1);
             } else if (d == full) {  // max digit + carry: digit 0, carry out
                 carry = 1;
                 key = skip_key;
             } else {  // negative digit, magnitude 2^c - d
                 carry = 1;
-                key = b * MSM_NB_TOTAL + w * MSM_BPW + (full - d - 1);
+                key = b * nb_local + ((uint32_t)w - w_lo) * MSM_BPW +
+                      (full - d - 1);
                 val |= 0x80000000u;
             }
-            const uint64_t slot = ((uint64_t)b * MSM_NWIN + w) * n + i;
+            if ((uint32_t)w < w_lo || (uint32_t)w >= w_lo + w_cnt) continue;
+            const uint64_t slot =
+                ((uint64_t)b * w_cnt + ((uint32_t)w - w_lo)) * n + i;
             keys[slot] = key;
             vals[slot] = val;
         }
@@ -266,9 +276,9 @@ __global__ __launch_bounds__(WSUM_THREADS, 2) void k_window_sum(
 
 // ---- host orchestration ---------------------------------------------------
 static int ensure_msm_scratch(MsmSlot& ds, hipStream_t stream, uint64_t n,
-                              uint32_t nbatch) {
-    const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
-    const uint64_t nbt = (uint64_t)nbatch * MSM_NB_TOTAL;
+                              uint32_t nbatch, uint32_t w_cnt) {
+    const uint64_t ent = (uint64_t)nbatch * w_cnt * n;
+    const uint64_t nbt = (uint64_t)nbatch * w_cnt * MSM_BPW;
     if (ds.ent_cap < ent) {
         // free-and-null before reallocating: a failed hipMalloc below must
         // not leave dangling pointers with a stale capacity (the caps are
@@ -298,7 +308,7 @@ static int ensure_msm_scratch(MsmSlot& ds, hipStream_t stream, uint64_t n,
         HIP_TRY(hipMalloc(&ds.d_offsets, (nbt + 1) * 4));
         HIP_TRY(hipMalloc(&ds.d_buckets, nbt * sizeof(g1_jac)));
         HIP_TRY(hipMalloc(&ds.d_red,
-                          (nbt / MSM_CHUNK + nbatch * MSM_NWIN) * sizeof(g1_jac)));
+                          (nbt / MSM_CHUNK + nbatch * w_cnt) * sizeof(g1_jac)));
         ds.nb_cap = nbt;
     }
     size_t sort_need = 0;
@@ -325,6 +335,19 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                            uint32_t nbatch, uint64_t n, uint32_t flags,
                            g1_jac* winsums_host, double* stage_ms, bool sync,
                            int slot) {
+    return msm_batch_windows_device(ctx, dev, d_bases, d_scalars, nbatch, n,
+                                    flags, 0, MSM_NWIN, winsums_host,
+                                    stage_ms, sync, slot);
+}
+
+// window-ranged pipeline: nwin = w_cnt windows starting at w_lo; winsums_host
+// receives nbatch * w_cnt Jacobian sums (full-range callers pass 0, NWIN).
+int msm_batch_windows_device(spectre_gpu_ctx* ctx, int dev,
+                             const g1_affine* d_bases,
+                             const uint8_t* d_scalars, uint32_t nbatch,
+                             uint64_t n, uint32_t flags, uint32_t w_lo,
+                             uint32_t w_cnt, g1_jac* winsums_host,
+                             double* stage_ms, bool sync, int slot) {
     DeviceState& dstate = ctx->devs[dev];
     HIP_TRY(hipSetDevice(dstate.device_id));
     if (slot < 0 || slot > 2) {
@@ -351,21 +374,27 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
     // nbatch*NWIN*n == 2^32 the final offset off[nb_total] wraps to 0 and the
     // result is silently wrong. Reject; callers split such jobs (e.g. batch
     // 32 x 2^23 -> 2 x 16).
-    if ((uint64_t)nbatch * MSM_NWIN * n >= (1ull << 32)) {
-        set_err("msm: nbatch*%d*n = %llu exceeds 2^32 sort-entry limit "
+
+    if (w_lo >= MSM_NWIN || w_cnt == 0 || w_lo + w_cnt > MSM_NWIN) {
+        set_err("msm: window range [%u, %u+%u) out of [0,%d)", w_lo, w_lo,
+                w_cnt, MSM_NWIN);
+        return -3;
+    }
+    if ((uint64_t)nbatch * w_cnt * n >= (1ull << 32)) {
+        set_err("msm: nbatch*%u*n = %llu exceeds 2^32 sort-entry limit "
                 "(split the batch or shard the MSM)",
-                MSM_NWIN, (unsigned long long)((uint64_t)nbatch * MSM_NWIN * n));
+                w_cnt, (unsigned long long)((uint64_t)nbatch * w_cnt * n));
         return -3;
     }
     if (n == 0) {
-        for (uint32_t w = 0; w < nbatch * MSM_NWIN; w++)
+        for (uint32_t w = 0; w < nbatch * w_cnt; w++)
             g1j_set_inf(winsums_host[w]);
         return 0;
     }
-    int rc = ensure_msm_scratch(ds, ds.stream, n, nbatch);
+    int rc = ensure_msm_scratch(ds, ds.stream, n, nbatch, w_cnt);
     if (rc) return rc;
-    const uint64_t ent = (uint64_t)nbatch * MSM_NWIN * n;
-    const uint32_t nbt = nbatch * MSM_NB_TOTAL;
+    const uint64_t ent = (uint64_t)nbatch * w_cnt * n;
+    const uint32_t nbt = nbatch * w_cnt * MSM_BPW;
     const int canonical = (flags & SPECTRE_SCALARS_CANONICAL) ? 1 : 0;
     hipStream_t st = ds.stream;
 
@@ -380,8 +409,8 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
     STAMP(0);
     uint32_t blocks = (uint32_t)((n + THREADS - 1) / THREADS);
     hipLaunchKernelGGL(k_msm_digits, dim3(blocks), dim3(THREADS), 0, st,
-                       d_scalars, n, nbatch, canonical, ds.d_keys_in,
-                       ds.d_vals_in);
+                       d_scalars, n, nbatch, canonical, w_lo, w_cnt,
+                       ds.d_keys_in, ds.d_vals_in);
     STAMP(1);
     size_t tmp = ds.sort_tmp_cap;
     (void)hipcub::DeviceRadixSort::SortPairs(
@@ -408,10 +437,10 @@ int msm_batch_shard_device(spectre_gpu_ctx* ctx, int dev,
                        dim3((nchunks + THREADS - 1) / THREADS), dim3(THREADS),
                        0, st, ds.d_buckets, nchunks, red0);
     STAMP(5);
-    hipLaunchKernelGGL(k_window_sum, dim3(nbatch * MSM_NWIN),
+    hipLaunchKernelGGL(k_window_sum, dim3(nbatch * w_cnt),
                        dim3(WSUM_THREADS), 0, st, red0, red1);
     STAMP(6);
-    const uint32_t nw = nbatch * MSM_NWIN;
+    const uint32_t nw = nbatch * w_cnt;
     if (ds.h_wins_cap < nw) {
         if (ds.h_wins) (void)hipHostFree(ds.h_wins);
         ds.h_wins_cap = 0;

@@ -82,6 +82,20 @@ def load_library() -> ctypes.CDLL:
     ]
     lib.spectre_gpu_msm_slot_wait.restype = c.c_int
     lib.spectre_gpu_msm_slot_wait.argtypes = [c.c_void_p, c.c_int, c.c_int]
+    lib.spectre_gpu_msm_g1_shard_windows_device.restype = c.c_int
+    lib.spectre_gpu_msm_g1_shard_windows_device.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
+        c.c_uint32, c.c_uint32, c.c_void_p,
+    ]
+    lib.spectre_gpu_msm_g1_shard_windows_device_async.restype = c.c_int
+    lib.spectre_gpu_msm_g1_shard_windows_device_async.argtypes = [
+        c.c_void_p, c.c_int, c.c_void_p, c.c_void_p, c.c_uint64, c.c_uint32,
+        c.c_uint32, c.c_uint32, c.c_void_p, c.POINTER(c.c_int),
+    ]
+    lib.spectre_gpu_msm_g1_combine_windows.restype = c.c_int
+    lib.spectre_gpu_msm_g1_combine_windows.argtypes = [
+        c.c_void_p, c.c_uint32, c.c_void_p,
+    ]
     lib.spectre_gpu_fr_gate_eval.restype = c.c_int
     lib.spectre_gpu_fr_gate_eval.argtypes = [
         c.c_void_p, c.c_int, c.POINTER(c.c_void_p), c.c_uint32, c.c_void_p,
@@ -123,6 +137,20 @@ def load_library() -> ctypes.CDLL:
 
 def version() -> str:
     return load_library().spectre_gpu_version().decode()
+
+
+def combine_window_partials(partials: bytes, nshards: int) -> bytes:
+    """Host-only: concatenated rank-ordered DISJOINT window slices (shard i =
+    NUM_WINDOWS/nshards consecutive windows) -> affine result."""
+    lib = load_library()
+    assert len(partials) == NUM_WINDOWS * 96
+    buf = (ctypes.c_uint8 * len(partials)).from_buffer_copy(partials)
+    out = (ctypes.c_uint8 * 64)()
+    rc = lib.spectre_gpu_msm_g1_combine_windows(buf, nshards, out)
+    if rc != 0:
+        raise RuntimeError(f"combine_windows failed rc={rc}: "
+                           f"{lib.spectre_gpu_last_error().decode()}")
+    return bytes(out)
 
 
 def combine_partials(partials: bytes, nshards: int) -> bytes:
@@ -264,6 +292,34 @@ class SpectreGpu:
     def msm_slot_wait(self, slot: int, dev: int = 0) -> None:
         rc = self._lib.spectre_gpu_msm_slot_wait(self._ctx, dev, slot)
         self._check(rc, "msm_slot_wait")
+
+    def msm_shard_windows_device(self, d_bases: int, d_scalars: int, n: int,
+                                 w_lo: int, w_cnt: int,
+                                 canonical: bool = True,
+                                 dev: int = 0) -> bytes:
+        """Window-sharded shard: this rank's w_cnt windows over ALL n
+        points (disjoint across ranks; exchange = pure allgather)."""
+        out = (ctypes.c_uint8 * (96 * w_cnt))()
+        rc = self._lib.spectre_gpu_msm_g1_shard_windows_device(
+            self._ctx, dev, ctypes.c_void_p(d_bases),
+            ctypes.c_void_p(d_scalars), n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY,
+            w_lo, w_cnt, out)
+        self._check(rc, "msm_g1_shard_windows_device")
+        return bytes(out)
+
+    def msm_shard_windows_device_async(self, d_bases: int, d_scalars: int,
+                                       n: int, w_lo: int, w_cnt: int,
+                                       canonical: bool = True, dev: int = 0):
+        out = (ctypes.c_uint8 * (96 * w_cnt))()
+        slot = ctypes.c_int(-1)
+        rc = self._lib.spectre_gpu_msm_g1_shard_windows_device_async(
+            self._ctx, dev, ctypes.c_void_p(d_bases),
+            ctypes.c_void_p(d_scalars), n,
+            SCALARS_CANONICAL if canonical else SCALARS_MONTGOMERY,
+            w_lo, w_cnt, out, ctypes.byref(slot))
+        self._check(rc, "msm_g1_shard_windows_device_async")
+        return out, slot.value
 
     def msm_shard_device_timed(self, d_bases: int, d_scalars: int, n: int,
                                canonical: bool = True, dev: int = 0):

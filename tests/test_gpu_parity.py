@@ -163,6 +163,35 @@ def test_msm_two_device_in_library(oracle):
         g.close()
 
 
+def test_msm_window_shard_combine_matches_direct(gpu, oracle):
+    """Window-sharded shards (disjoint window ranges over ALL points,
+    concatenated in rank order) must equal the direct result bit-for-bit —
+    the default N>1 exchange (pure allgather, no reduction)."""
+    from spectre_amd import ffi
+    n = 1 << 13
+    sc, bs = oracle.gen_msm_inputs(n, 64, fast=True)
+    direct = gpu.msm(bs, sc, n)
+    d_b = gpu.malloc(64 * n)
+    d_s = gpu.malloc(32 * n)
+    gpu.upload(d_b, bs)
+    gpu.upload(d_s, sc)
+    for world in (2, 4, 8):
+        w_cnt = ffi.NUM_WINDOWS // world
+        blob = b"".join(
+            gpu.msm_shard_windows_device(d_b, d_s, n, r * w_cnt, w_cnt)
+            for r in range(world))
+        assert ffi.combine_window_partials(blob, world) == direct, world
+    # async variant, interleaved across two ranks' window sets
+    w_cnt = ffi.NUM_WINDOWS // 2
+    b0, s0 = gpu.msm_shard_windows_device_async(d_b, d_s, n, 0, w_cnt)
+    b1, s1 = gpu.msm_shard_windows_device_async(d_b, d_s, n, w_cnt, w_cnt)
+    gpu.msm_slot_wait(s0)
+    gpu.msm_slot_wait(s1)
+    assert ffi.combine_window_partials(bytes(b0) + bytes(b1), 2) == direct
+    gpu.free(d_b)
+    gpu.free(d_s)
+
+
 def test_msm_shard_combine_matches_direct(gpu, oracle):
     """Two shards on one device + host combine == unsharded result — the
     exact exchange the multi-GPU path performs (bit-identical by affine
