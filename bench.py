@@ -255,7 +255,8 @@ def main():
                 "workload": "msm_g1_2pow20",
                 "n": N,
                 "seed": SEED,
-                "parallelism": f"shard{n_gpus}" + (f"+{backend}" if world > 1 else ""),
+                "parallelism": (f"shard{n_gpus}:{shard_mode}" +
+                                (f"+{backend}" if world > 1 else "")),
                 "pipeline_depth": (max(2, int(os.environ.get(
                     "SPECTRE_PIPE_SLOTS", "3"))) if pipelined else 1),
                 "scalars": "canonical",
