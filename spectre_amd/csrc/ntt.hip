@@ -646,6 +646,19 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         ff_inv<Fr>(scale, nm);
     }
     hipStream_t st = ds.stream;
+    // SPECTRE_NTT_TDIV=1|2|4: launch L/TDIV threads per tile instead of L.
+    // Radix-2 butterflies occupy only L/2 threads and radix-4 only L/4 of
+    // an L-thread block; smaller blocks let several blocks co-reside per CU
+    // (LDS permitting) and auto-enable radix-4 (its L >= 4*threads gate).
+    static const uint32_t kTDiv = []() {
+        const char* e = getenv("SPECTRE_NTT_TDIV");
+        int v = e ? atoi(e) : 1;
+        return (uint32_t)(v == 2 || v == 4 ? v : 1);
+    }();
+    auto tdiv = [](uint32_t L) {
+        uint32_t t = L / kTDiv;
+        return t < 64 ? 64u : (t > NTT_THREADS ? (uint32_t)NTT_THREADS : t);
+    };
     const fp256* fwd_cT1 = (coset_gen && !inverse) ? cT1 : nullptr;
     const fp256* fwd_cT2 = (coset_gen && !inverse) ? cT2 : nullptr;
     const fp256* inv_cT1 = (coset_gen && inverse) ? cT1 : nullptr;
@@ -658,9 +671,7 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         ff_set_one<Fr>(one);
         const fp256* T1 = plan->twB;
         const fp256* T2 = plan->twB + t1n;
-        auto tdim = [](uint32_t L) {
-            return L < NTT_THREADS ? (L < 64 ? 64u : L) : (uint32_t)NTT_THREADS;
-        };
+        auto tdim = tdiv;
         // radix-4 instantiation only when every thread gets a quad
         auto axis = [](uint32_t L, uint32_t threads) {
             return (!kForceRadix2 && L >= 4 * threads) ? k_ntt_axis<true>
@@ -680,8 +691,8 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
                            scale, inverse ? 1 : 0, kC, 0, /*post_mode=*/0, 0,
                            0, /*final_scatter=*/1, A, B);
     } else if (plan->k1 > 0) {
-        const uint32_t tc = n1 < NTT_THREADS ? (n1 < 64 ? 64 : n1) : NTT_THREADS;
-        const uint32_t tr = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
+        const uint32_t tc = tdiv(n1);
+        const uint32_t tr = tdiv(n2);
         hipLaunchKernelGGL((!kForceRadix2 && n1 >= 4 * tc) ? k_ntt_col<true>
                                                            : k_ntt_col<false>,
                            dim3(n2), dim3(tc),
@@ -700,7 +711,7 @@ int ntt_device(spectre_gpu_ctx* ctx, int dev, fp256* d_data, uint32_t log_n,
         // single pass; forward coset applies on load, inverse on store
         const fp256* cc1 = coset_gen ? cT1 : nullptr;
         const fp256* cc2 = coset_gen ? cT2 : nullptr;
-        const uint32_t t1p = n2 < NTT_THREADS ? (n2 < 64 ? 64 : n2) : NTT_THREADS;
+        const uint32_t t1p = tdiv(n2);
         hipLaunchKernelGGL((!kForceRadix2 && n2 >= 4 * t1p) ? k_ntt_row<true>
                                                             : k_ntt_row<false>,
                            dim3(1), dim3(t1p),
