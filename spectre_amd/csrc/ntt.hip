@@ -34,7 +34,9 @@
 // 1024-thread LDS kernels ~7% (2^23 fwd 2.59 -> 2.77 ms — register pressure
 // at the 4-waves/SIMD occupancy these kernels need), while it wins ~28% in
 // the 256-thread MSM kernels. Per-TU choice, measured r2.
-#define SPECTRE_NO_ASM_MUL 1
+#ifndef SPECTRE_NTT_ASMMUL  // variant: -DSPECTRE_NTT_ASMMUL re-tests the
+#define SPECTRE_NO_ASM_MUL 1  // asm multiply under new block geometries
+#endif
 #include "internal.hpp"
 
 #define THREADS 256       // pow-table builder
