@@ -63,6 +63,24 @@ __global__ __launch_bounds__(256, 2) void k_mul_cols_chain(
     out[t].X = a;
 }
 
+// two INDEPENDENT product-path mul chains per thread: if this runs much
+// faster than 2x the single-chain time, the single chain is stall-bound
+// (dependent-mad latency) and a dual-accumulator column schedule would pay;
+// if it matches, the mul is issue-bound and deeper ILP buys nothing.
+__global__ __launch_bounds__(256, 2) void k_mul2_chain(const g1_affine* pts,
+                                                       g1_jac* out, int nwork) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= nwork) return;
+    fp256 a = pts[t % 1024].x, b = pts[t % 1024].y;
+    fp256 a2 = pts[(t + 7) % 1024].y, b2 = pts[(t + 7) % 1024].x;
+    for (int i = 0; i < ITERS * 8; i++) {
+        ff_mul<Fq>(a, a, b);
+        ff_mul<Fq>(a2, a2, b2);
+    }
+    out[t].X = a;
+    out[t].Y = a2;
+}
+
 // correctness: ff_mul_cols must equal ff_mul bit-for-bit on pseudorandom
 // reduced inputs (both Fq and Fr), chained so errors compound and surface.
 __global__ void k_mul_cols_check(uint32_t* bad, int nwork) {
@@ -143,6 +161,10 @@ int main() {
         double msc = time_kernel(k_mul_cols_chain, pts, out, nwork, nwork / 256);
         printf("asm ff_mul chain: %.3f ms  (%.1f M mul/s, %.2fx vs C)\n",
                msc, (double)nwork * ITERS * 16 / msc / 1e3, msm / msc);
+        double ms2 = time_kernel(k_mul2_chain, pts, out, nwork, nwork / 256);
+        printf("asm ff_mul 2-indep chains: %.3f ms  (%.1f M mul/s total; "
+               "stall-bound if >> the 1-chain rate)\n",
+               ms2, (double)nwork * ITERS * 16 / ms2 / 1e3);
         (void)hipFree(bad);
     }
     // raw mad64 rate
