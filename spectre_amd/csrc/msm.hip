@@ -36,26 +36,6 @@
 // to modular arithmetic.
 #include "internal.hpp"
 #include <hipcub/hipcub.hpp>
-#include <rocprim/device/device_radix_sort.hpp>
-
-// Optional 10-bit-digit onesweep (2 iterations for the 20-bit bucket keys
-// vs 3 with the default 8-bit config). 256-thread sort blocks keep the
-// match-rank LDS in budget at 1024 bins; the call's rc IS checked — the
-// first attempt (1024-thread blocks) failed at run time and, unchecked,
-// silently skipped the sort. Enabled with SPECTRE_SORT10=1 until proven.
-using Sort10Config = rocprim::radix_sort_config<
-    rocprim::default_config, rocprim::default_config,
-    rocprim::radix_sort_onesweep_config<
-        rocprim::kernel_config<256, 12>, rocprim::kernel_config<256, 15>, 10,
-        rocprim::block_radix_rank_algorithm::match>>;
-
-static bool sort10_enabled() {
-    static const bool v = []() {
-        const char* e = getenv("SPECTRE_SORT10");
-        return e && e[0] == '1';
-    }();
-    return v;
-}
 
 #define THREADS 256
 // Point-arithmetic kernels hold a 24-VGPR Jacobian accumulator plus formula
@@ -433,15 +413,9 @@ int msm_batch_windows_device(spectre_gpu_ctx* ctx, int dev,
                        ds.d_keys_in, ds.d_vals_in);
     STAMP(1);
     size_t tmp = ds.sort_tmp_cap;
-    if (sort10_enabled()) {
-        HIP_TRY(rocprim::radix_sort_pairs<Sort10Config>(
-            ds.d_sort_tmp, tmp, ds.d_keys_in, ds.d_keys_out, ds.d_vals_in,
-            ds.d_vals_out, ent, 0, (unsigned)end_bit_for(nbt), st));
-    } else {
-        (void)hipcub::DeviceRadixSort::SortPairs(
-            ds.d_sort_tmp, tmp, ds.d_keys_in, ds.d_keys_out, ds.d_vals_in,
-            ds.d_vals_out, (int64_t)ent, 0, end_bit_for(nbt), st);
-    }
+    (void)hipcub::DeviceRadixSort::SortPairs(
+        ds.d_sort_tmp, tmp, ds.d_keys_in, ds.d_keys_out, ds.d_vals_in,
+        ds.d_vals_out, (int64_t)ent, 0, end_bit_for(nbt), st);
     STAMP(2);
     hipLaunchKernelGGL(k_bucket_offsets,
                        dim3((nbt + 1 + THREADS - 1) / THREADS), dim3(THREADS),
