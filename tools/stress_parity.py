@@ -35,7 +35,8 @@ def main():
         return w
 
     while time.time() - t0 < budget_s:
-        kind = rng.choice(["msm", "msm_batch", "msm_shard", "ntt"])
+        kind = rng.choice(["msm", "msm_batch", "msm_shard", "msm_windows",
+                           "ntt", "gate"])
         if kind == "msm":
             n = rng.randrange(1, 1 << rng.randrange(4, 17))
             sc, bs = oracle.gen_msm_inputs(n, rng.randrange(1 << 30), fast=True)
@@ -75,6 +76,85 @@ def main():
                 gpu.free(d_b)
                 gpu.free(d_s)
             assert ffi.combine_partials(parts, nsh) == want, (kind, n, nsh)
+        elif kind == "msm_windows":
+            # window-sharded shards (disjoint window ranges, rank order)
+            n = rng.randrange(64, 1 << 13)
+            nsh = rng.choice([2, 4, 8])
+            sc, bs = oracle.gen_msm_inputs(n, rng.randrange(1 << 30), fast=True)
+            want = gpu.msm(bs, sc, n)
+            d_b = gpu.malloc(64 * n)
+            d_s = gpu.malloc(32 * n)
+            gpu.upload(d_b, bs)
+            gpu.upload(d_s, sc)
+            w_cnt = ffi.NUM_WINDOWS // nsh
+            blob = b"".join(
+                gpu.msm_shard_windows_device(d_b, d_s, n, r * w_cnt, w_cnt)
+                for r in range(nsh))
+            gpu.free(d_b)
+            gpu.free(d_s)
+            assert ffi.combine_window_partials(blob, nsh) == want, (kind, n, nsh)
+        elif kind == "gate":
+            # random well-formed gate program vs bigint evaluation
+            log_n = rng.randrange(4, 11)
+            n = 1 << log_n
+            ncols = rng.randrange(1, 4)
+            cols_int = [[rng.randrange(R) for _ in range(n)]
+                        for _ in range(ncols)]
+            consts = [rng.randrange(R) for _ in range(2)]
+            rot_scale = rng.choice([1, 2, 4])
+            prog, depth = [], 0
+            while len(prog) < rng.randrange(3, 24) or depth != 1:
+                if depth >= 2 and (depth >= 7 or rng.random() < 0.5):
+                    prog.append((rng.choice([gpu.GATE_ADD, gpu.GATE_SUB,
+                                             gpu.GATE_MUL]), 0, 0))
+                    depth -= 1
+                elif depth >= 1 and rng.random() < 0.1:
+                    prog.append((gpu.GATE_NEG, 0, 0))
+                elif rng.random() < 0.7:
+                    max_rot = max(1, (n - 1) // rot_scale)
+                    r_lo = -min(3, max_rot)
+                    r_hi = min(3, max_rot)
+                    prog.append((gpu.GATE_COL, rng.randrange(ncols),
+                                 rng.randrange(r_lo, r_hi + 1)))
+                    depth += 1
+                else:
+                    prog.append((gpu.GATE_CONST, rng.randrange(2), 0))
+                    depth += 1
+            fr_vec = lambda ints: b"".join(
+                oracle.fr_from_canonical(v.to_bytes(32, "little"))
+                for v in ints)
+            d_cols = []
+            for ci in cols_int:
+                d = gpu.malloc(32 * n)
+                gpu.upload(d, fr_vec(ci))
+                d_cols.append(d)
+            d_out = gpu.malloc(32 * n)
+            gpu.gate_eval(d_cols, fr_vec(consts), prog, n,
+                          rot_scale=rot_scale, d_out=d_out)
+            raw = bytes(gpu.download(d_out, 32 * n))
+            got = [int.from_bytes(
+                oracle.fr_to_canonical(raw[32 * i:32 * (i + 1)]), "little")
+                for i in range(n)]
+            want = []
+            for row in range(n):
+                st = []
+                for op, a_, b_ in prog:
+                    if op == gpu.GATE_COL:
+                        st.append(cols_int[a_][(row + b_ * rot_scale) % n])
+                    elif op == gpu.GATE_CONST:
+                        st.append(consts[a_])
+                    elif op == gpu.GATE_NEG:
+                        st[-1] = (-st[-1]) % R
+                    else:
+                        rhs = st.pop()
+                        lhs = st.pop()
+                        st.append((lhs + rhs) % R if op == gpu.GATE_ADD else
+                                  (lhs - rhs) % R if op == gpu.GATE_SUB else
+                                  lhs * rhs % R)
+                want.append(st[0])
+            for d in d_cols + [d_out]:
+                gpu.free(d)
+            assert got == want, (kind, log_n, len(prog))
         else:
             log_n = rng.randrange(1, 19)
             n = 1 << log_n
