@@ -174,7 +174,7 @@ template <class C> FF_HD void ff_mul_cios(fp256& o, const fp256& a, const fp256&
     for (int i = 0; i < 8; i++) o.l[i] = t[i];
     ff_cond_sub_mod<C>(o, t[8]);
 }
-// ---- device multiply: hand-scheduled column Montgomery (ff_asm.hpp) ------
+// ---- device multiply: hand-scheduled column Montgomery --------------------
 // On gfx950 the asm column form runs at ~135 G Fq-mul/s vs ~103 for the
 // compiled CIOS (measured, tools/microbench.hip) and is bit-identical
 // (256K-lane chained check + the whole GPU parity suite). Define
