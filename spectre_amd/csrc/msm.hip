@@ -44,6 +44,15 @@
 // latency hiding comes from the serial-add structure, not occupancy
 // (measured: 1 add-chain/thread already saturates the VALU issue pipe).
 #define PT_KERNEL __global__ __launch_bounds__(THREADS, 2)
+// The accumulate/fix kernels have no LDS or barriers, so their block size
+// only sets the residency granule: the asm multiply brought them to 84
+// VGPRs (5 waves/SIMD by registers), but 256-thread blocks stack in
+// 2-wave/SIMD steps (cap 4). 128-thread blocks allow 5 (SQ counters show
+// the waves memory-parked 24% of the time — more waves hide it).
+#ifndef MSM_ACC_THREADS
+#define MSM_ACC_THREADS 128
+#endif
+#define ACC_KERNEL __global__ __launch_bounds__(MSM_ACC_THREADS, 2)
 
 // ---- kernel 1: signed window decomposition --------------------------------
 // w_lo/w_cnt: emit only windows [w_lo, w_lo+w_cnt) — window-sharded
@@ -122,7 +131,7 @@ __global__ void k_bucket_offsets(const uint32_t* __restrict__ keys,
 // AND end strictly inside the range write their bucket directly (exclusive);
 // the first and last (potentially thread-spanning) runs go to side arrays
 // keyed by bucket, merged by k_bucket_fix.
-PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
+ACC_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
                             const uint32_t* __restrict__ keys,
                             const uint32_t* __restrict__ vals,
                             const g1_affine* __restrict__ bases,
@@ -180,7 +189,7 @@ PT_KERNEL void k_bucket_acc(const uint32_t* __restrict__ off,
 
 // merge boundary partials: bucket b's segment [s,e) spans threads ts..te;
 // interior-only buckets were already written by their exclusive thread.
-PT_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
+ACC_KERNEL void k_bucket_fix(const uint32_t* __restrict__ off,
                             const uint32_t* __restrict__ firstK,
                             const g1_jac* __restrict__ firstP,
                             const uint32_t* __restrict__ lastK,
@@ -422,13 +431,17 @@ int msm_batch_windows_device(spectre_gpu_ctx* ctx, int dev,
                        0, st, ds.d_keys_out, ent, nbt, ds.d_offsets);
     STAMP(3);
     const uint32_t nt_acc = (uint32_t)((ent + MSM_ACC_E - 1) / MSM_ACC_E);
-    hipLaunchKernelGGL(k_bucket_acc, dim3((nt_acc + THREADS - 1) / THREADS),
-                       dim3(THREADS), 0, st, ds.d_offsets, ds.d_keys_out,
-                       ds.d_vals_out, d_bases, nbt, ds.d_buckets, ds.d_firstK,
-                       ds.d_firstP, ds.d_lastK, ds.d_lastP);
-    hipLaunchKernelGGL(k_bucket_fix, dim3((nbt + THREADS - 1) / THREADS),
-                       dim3(THREADS), 0, st, ds.d_offsets, ds.d_firstK,
-                       ds.d_firstP, ds.d_lastK, ds.d_lastP, nbt, ds.d_buckets);
+    hipLaunchKernelGGL(k_bucket_acc,
+                       dim3((nt_acc + MSM_ACC_THREADS - 1) / MSM_ACC_THREADS),
+                       dim3(MSM_ACC_THREADS), 0, st, ds.d_offsets,
+                       ds.d_keys_out, ds.d_vals_out, d_bases, nbt,
+                       ds.d_buckets, ds.d_firstK, ds.d_firstP, ds.d_lastK,
+                       ds.d_lastP);
+    hipLaunchKernelGGL(k_bucket_fix,
+                       dim3((nbt + MSM_ACC_THREADS - 1) / MSM_ACC_THREADS),
+                       dim3(MSM_ACC_THREADS), 0, st, ds.d_offsets,
+                       ds.d_firstK, ds.d_firstP, ds.d_lastK, ds.d_lastP, nbt,
+                       ds.d_buckets);
     STAMP(4);
     const uint32_t nchunks = nbt / MSM_CHUNK;
     g1_jac* red0 = ds.d_red;
