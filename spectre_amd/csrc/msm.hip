@@ -45,12 +45,12 @@
 // (measured: 1 add-chain/thread already saturates the VALU issue pipe).
 #define PT_KERNEL __global__ __launch_bounds__(THREADS, 2)
 // The accumulate/fix kernels have no LDS or barriers, so their block size
-// only sets the residency granule: the asm multiply brought them to 84
-// VGPRs (5 waves/SIMD by registers), but 256-thread blocks stack in
-// 2-wave/SIMD steps (cap 4). 128-thread blocks allow 5 (SQ counters show
-// the waves memory-parked 24% of the time — more waves hide it).
+// only sets the residency granule. MEASURED (r2): 64/128/256-thread blocks
+// are within run-to-run noise (acc 2.26-2.29 ms) — the 24% memory-parked
+// wave time (SQ_WAIT_ANY) does not yield to the extra wave 128-thread
+// blocks admit (84 VGPRs -> 5 waves/SIMD vs 4). Kept tunable.
 #ifndef MSM_ACC_THREADS
-#define MSM_ACC_THREADS 128
+#define MSM_ACC_THREADS 256
 #endif
 #define ACC_KERNEL __global__ __launch_bounds__(MSM_ACC_THREADS, 2)
 
