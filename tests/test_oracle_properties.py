@@ -83,3 +83,37 @@ def test_msm_fast_gen_on_curve(oracle):
     _, bs = oracle.gen_msm_inputs(64, 30, fast=True)
     for i in range(64):
         assert oracle.g1_is_on_curve(bs[64 * i:64 * i + 64])
+
+
+def test_g2_group_properties(oracle):
+    """Randomized G2 algebra: commutativity, associativity, doubling
+    consistency, scalar-mul additivity, negation — independent of the
+    committed fixtures (which pin values; these pin structure)."""
+    import random
+    rng = random.Random(90210)
+    R = 21888242871839275222246405745257275088548364400416034343698204186575808495617
+    import json
+    import os
+    g2fix = json.load(open(os.path.join(
+        os.path.dirname(__file__), "golden", "g2.json")))
+    G = bytes.fromhex(g2fix["mul_cases"][1]["mul"])  # 1*G = generator
+    assert oracle.g2_is_on_curve(G)
+    for _ in range(6):
+        a, b = rng.randrange(R), rng.randrange(R)
+        P = oracle.g2_mul(G, a.to_bytes(32, "little"))
+        Q = oracle.g2_mul(G, b.to_bytes(32, "little"))
+        assert oracle.g2_is_on_curve(P) and oracle.g2_is_on_curve(Q)
+        # commutativity
+        assert oracle.g2_add(P, Q) == oracle.g2_add(Q, P)
+        # additivity: aG + bG == (a+b)G
+        S = oracle.g2_mul(G, ((a + b) % R).to_bytes(32, "little"))
+        assert oracle.g2_add(P, Q) == S
+        # doubling == adding to itself == 2k*G
+        D = oracle.g2_add(P, P)
+        assert D == oracle.g2_mul(G, (2 * a % R).to_bytes(32, "little"))
+        # negation cancels
+        assert oracle.g2_add(P, oracle.g2_neg(P)) == bytes(128)
+        # associativity with a third point
+        T = oracle.g2_mul(G, rng.randrange(R).to_bytes(32, "little"))
+        assert (oracle.g2_add(oracle.g2_add(P, Q), T)
+                == oracle.g2_add(P, oracle.g2_add(Q, T)))
