@@ -90,6 +90,29 @@ def test_allgather_combine_windows_gloo(oracle, golden):
     assert results[0] == results[1] == want.hex()
 
 
+def test_allgather_combine_windows_gloo_world4(oracle, golden):
+    """Window exchange at world 4 (each rank 4 windows) — the shape the
+    round-end 8-GPU scaling run exercises, CPU-validated with gloo."""
+    g1 = golden("g1.json")
+    pts = [c["mul"] for c in g1["mul_cases"]
+           if bytes.fromhex(c["mul"]) != bytes(64)][:4]
+    world = 4
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_windows, args=(world, 29579, pts, results), nprocs=world,
+             join=True)
+    from spectre_amd import ffi
+    want = None
+    wb = ffi.WINDOW_BITS
+    per = ffi.NUM_WINDOWS // world
+    for r in range(world):
+        term = oracle.g1_mul(bytes.fromhex(pts[r]),
+                             (1 << (wb * per * r)).to_bytes(32, "little"))
+        want = term if want is None else oracle.g1_add(want, term)
+    for r in range(world):
+        assert results[r] == want.hex(), r
+
+
 def test_allgather_combine_gloo(oracle, golden):
     g1 = golden("g1.json")
     pts = [c["mul"] for c in g1["mul_cases"]
